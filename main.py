@@ -1,0 +1,197 @@
+"""Train CLI — API-compatible with the reference ``main.py`` (same flags:
+--run/--experiment/--disable-logging/--render/--environment/--cpus,
+reference main.py:113-125; same hyperparameter defaults, main.py:147-160;
+same MLflow checkpoint/resume layout, main.py:28-51) plus MI355X flags:
+--gpus (per-GPU process data parallelism over RCCL/xGMI), --device,
+--batch-size/--buffer-size overrides, --learn-alpha, --reference-pi-loss.
+
+Structural fixes vs the reference: the multi-process fork
+happens BEFORE any heavy state is built, and the worker-count flags are
+typed ints.
+"""
+
+import logging
+from argparse import ArgumentParser, Namespace
+from pathlib import Path
+
+import torch
+
+from buffer.replay_buffer import ReplayBuffer
+from buffer.visual_replay_buffer import VisualReplayBuffer
+from networks.convolutional import VisualActor, VisualDoubleCritic
+from networks.linear import Actor, DoubleCritic
+from sac.algorithm import SAC
+from torch_actor_critic_amd import envs
+from torch_actor_critic_amd.optim import FlatAdam
+from torch_actor_critic_amd.parallel import comm
+from torch_actor_critic_amd.parallel.launch import gpu_fork
+from torch_actor_critic_amd.utils import checkpoint as ckpt
+
+logging.basicConfig(level=logging.INFO)
+logger = logging.getLogger(__name__)
+
+VISUAL_ENVS = ("DeepMindWallRunner-v0", "VisualCheetahRun-v0")
+
+
+def load_session(run_id: str, device: torch.device):
+    """Resume from an MLflow run (reference main.py:28-51)."""
+    sac_params = ckpt.get_run_params(run_id)
+    artifacts = Path("mlruns", "0", run_id, "artifacts")
+    actor = ckpt.load_model(str(artifacts / "actor")).to(device)
+    critic = ckpt.load_model(str(artifacts / "critic")).to(device)
+    auxiliaries = ckpt.load_state_dict(str(artifacts / "auxiliaries"))
+
+    pi_opt = FlatAdam(actor)
+    pi_opt.load_state_dict(auxiliaries["pi_opt"])
+    q_opt = FlatAdam(critic)
+    q_opt.load_state_dict(auxiliaries["q_opt"])
+    start_epoch = auxiliaries["epoch"]
+
+    # not SAC-constructor params (the reference forgets buffer_size and
+    # crashes on resume when it was logged — fixed here)
+    sac_params.pop("environment", None)
+    sac_params.pop("buffer_size", None)
+    sac_params = {
+        k: int(float(v)) if float(v).is_integer() else float(v)
+        for k, v in sac_params.items()
+    }
+    return actor, critic, pi_opt, q_opt, start_epoch, sac_params
+
+
+def init_session(environment: str, device: torch.device):
+    """Fresh nets + optimizers (reference main.py:54-97)."""
+    env = envs.make(environment)
+    act_dim = env.action_space.shape[0]
+    obs_dim = env.observation_space.shape[0]
+    act_limit = float(env.action_space.high.reshape(-1)[0])
+    hidden_sizes = [256, 256]
+
+    if environment in VISUAL_ENVS:
+        vis_dim = (3, 64, 64) if environment == "DeepMindWallRunner-v0" \
+            else (3, 84, 84)
+        actor = VisualActor(obs_dim, act_dim, vis_dim, hidden_sizes,
+                            act_limit)
+        critic = VisualDoubleCritic(obs_dim, act_dim, vis_dim, hidden_sizes)
+    else:
+        actor = Actor(obs_dim, act_dim, hidden_sizes, act_limit=act_limit)
+        critic = DoubleCritic(obs_dim, act_dim, hidden_sizes)
+    actor = actor.to(device)
+    critic = critic.to(device)
+
+    learning_rate = 3e-4
+    pi_opt = FlatAdam(actor, lr=learning_rate)
+    q_opt = FlatAdam(critic, lr=learning_rate)
+    return actor, critic, pi_opt, q_opt, 0
+
+
+def init_buffer(environment: str, size: int, device: torch.device):
+    """HBM-resident replay buffer (reference main.py:100-110)."""
+    env = envs.make(environment)
+    act_dim = env.action_space.shape[0]
+    obs_dim = env.observation_space.shape[0]
+    if environment in VISUAL_ENVS:
+        return VisualReplayBuffer(size, act_dim, device=device)
+    return ReplayBuffer(size, obs_dim, act_dim, device=device)
+
+
+def parse_arguments() -> Namespace:
+    parser = ArgumentParser("Soft Actor-Critic trainer (MI355X-native).")
+    parser.add_argument("--run", type=str, default=None,
+                        help="Resume from an existing mlflow run id")
+    parser.add_argument("--experiment", default="Default",
+                        help="Mlflow experiment name")
+    parser.add_argument("--disable-logging", dest="logging",
+                        action="store_false", help="Turn off logging")
+    parser.add_argument("--render", action="store_true",
+                        help="Enable environment rendering")
+    parser.add_argument("--environment", default="Humanoid-v2",
+                        help="Environment to use")
+    parser.add_argument("--cpus", type=int, default=1,
+                        help="Number of CPU data-parallel ranks")
+    parser.add_argument("--gpus", type=int, default=0,
+                        help="Number of GPUs (one process per MI355X)")
+    parser.add_argument("--device", default=None,
+                        help="Compute device (default: cuda if available)")
+    parser.add_argument("--buffer-size", type=int, default=int(1e6))
+    parser.add_argument("--batch-size", type=int, default=None,
+                        help="Override SAC batch size")
+    parser.add_argument("--epochs", type=int, default=None)
+    parser.add_argument("--steps-per-epoch", type=int, default=None)
+    parser.add_argument("--learn-alpha", action="store_true",
+                        help="Learned entropy temperature (extension)")
+    parser.add_argument("--reference-pi-loss", action="store_true",
+                        help="Reproduce the reference's next_state policy "
+                             "loss quirk (SURVEY.md Q2)")
+    parser.set_defaults(logging=True, render=False)
+    return parser.parse_args()
+
+
+def main():
+    args = parse_arguments()
+
+    # fork BEFORE building any state (fix of reference Q4, main.py:168)
+    n_workers = max(args.gpus, args.cpus)
+    gpu_fork(n_workers)
+    comm.init_distributed()
+
+    if args.device is not None:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device("cuda",
+                              comm.proc_id() % max(torch.cuda.device_count(), 1))
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+        torch.set_num_threads(2)
+
+    ckpt.set_experiment(args.experiment)
+    rank0 = comm.proc_id() == 0
+    if rank0 and args.logging:
+        if args.run is None:
+            ckpt.start_run()
+        else:
+            ckpt._ACTIVE_RUN = args.run  # resume into the same run dir
+
+    buffer = init_buffer(args.environment, args.buffer_size, device)
+
+    if args.run is not None:
+        actor, critic, pi_opt, q_opt, start_epoch, params = \
+            load_session(args.run, device)
+    else:
+        actor, critic, pi_opt, q_opt, start_epoch = \
+            init_session(args.environment, device)
+        params = dict(
+            alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+            epochs=1000, batch_size=64, steps_per_epoch=5000,
+            start_steps=1000, update_after=1000, update_every=50,
+            max_ep_len=5000, save_every=10,
+        )
+        if rank0 and args.logging:
+            ckpt.log_params(params)
+            ckpt.log_params({"environment": args.environment,
+                             "buffer_size": args.buffer_size})
+
+    if args.batch_size is not None:
+        params["batch_size"] = args.batch_size
+    if args.epochs is not None:
+        params["epochs"] = args.epochs
+    if args.steps_per_epoch is not None:
+        params["steps_per_epoch"] = args.steps_per_epoch
+
+    sac = SAC(**params, learn_alpha=args.learn_alpha,
+              reference_pi_loss=args.reference_pi_loss)
+    env = envs.make(args.environment)
+
+    try:
+        sac.train(
+            start_epoch=start_epoch, buffer=buffer, env=env, actor=actor,
+            critic=critic, pi_opt=pi_opt, q_opt=q_opt, render=args.render,
+            logging=args.logging,
+        )
+    finally:
+        if rank0 and args.logging:
+            ckpt.end_run()
+
+
+if __name__ == "__main__":
+    main()

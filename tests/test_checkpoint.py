@@ -1,0 +1,93 @@
+"""Checkpoint layout and resume tests (reference artifact layout:
+mlruns/0/<run_id>/artifacts/{actor,critic,auxiliaries} —
+sac/algorithm.py:164-180, main.py:28-51)."""
+
+import os
+
+import torch
+
+from networks.linear import Actor, DoubleCritic
+from torch_actor_critic_amd.optim import FlatAdam
+from torch_actor_critic_amd.utils import checkpoint as ckpt
+
+
+def _setup(tmp_path):
+    ckpt.set_tracking_dir(str(tmp_path / "mlruns"))
+    return ckpt.start_run()
+
+
+def test_artifact_layout(tmp_path):
+    run_id = _setup(tmp_path)
+    actor = Actor(4, 2, [8])
+    critic = DoubleCritic(4, 2, [8])
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+
+    ckpt.log_model(actor, "actor")
+    ckpt.log_model(critic, "critic")
+    ckpt.log_state_dict({"pi_opt": pi_opt.state_dict(),
+                         "q_opt": q_opt.state_dict(), "epoch": 7},
+                        "auxiliaries")
+    ckpt.log_params({"alpha": 0.2, "environment": "Pendulum-v1"})
+    ckpt.log_metrics({"reward": -1.0}, step=0)
+    ckpt.end_run()
+
+    base = tmp_path / "mlruns" / "0" / run_id
+    assert (base / "artifacts" / "actor" / "data" / "model.pth").exists()
+    assert (base / "artifacts" / "actor" / "MLmodel").exists()
+    assert (base / "artifacts" / "critic" / "data" / "model.pth").exists()
+    assert (base / "artifacts" / "auxiliaries" / "state_dict.pth").exists()
+    assert (base / "params" / "alpha").read_text() == "0.2"
+    assert "reward" in os.listdir(base / "metrics")
+
+
+def test_model_roundtrip(tmp_path):
+    run_id = _setup(tmp_path)
+    actor = Actor(4, 2, [8], act_limit=2.0)
+    ckpt.log_model(actor, "actor")
+    ckpt.end_run()
+
+    path = str(tmp_path / "mlruns" / "0" / run_id / "artifacts" / "actor")
+    loaded = ckpt.load_model(path)
+    obs = torch.randn(5, 4)
+    p1, _ = actor(obs, deterministic=True)
+    p2, _ = loaded(obs, deterministic=True)
+    assert torch.allclose(p1, p2)
+
+
+def test_resume_via_main_load_session(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    ckpt.set_tracking_dir("mlruns")
+    run_id = ckpt.start_run()
+
+    actor = Actor(3, 1, [8], act_limit=2.0)
+    critic = DoubleCritic(3, 1, [8])
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+    pi, logp = actor(torch.randn(4, 3))
+    (pi.sum() + logp.sum()).backward()
+    pi_opt.step()
+
+    ckpt.log_params({"alpha": 0.2, "gamma": 0.99, "polyak": 0.995,
+                     "reward_scale": 1.0, "epochs": 10, "batch_size": 64,
+                     "steps_per_epoch": 100, "start_steps": 10,
+                     "update_after": 10, "update_every": 5,
+                     "max_ep_len": 100, "save_every": 2,
+                     "environment": "Pendulum-v1", "buffer_size": 1000})
+    ckpt.log_model(actor, "actor")
+    ckpt.log_model(critic, "critic")
+    ckpt.log_state_dict({"pi_opt": pi_opt.state_dict(),
+                         "q_opt": q_opt.state_dict(), "epoch": 3},
+                        "auxiliaries")
+    ckpt.end_run()
+
+    import main as train_main
+    a2, c2, p2, q2, epoch, params = train_main.load_session(
+        run_id, torch.device("cpu"))
+    assert epoch == 3
+    assert params["alpha"] == 0.2
+    assert params["epochs"] == 10
+    assert "environment" not in params and "buffer_size" not in params
+    assert int(p2.step_t.item()) == 1
+    obs = torch.randn(2, 3)
+    o1, _ = actor(obs, deterministic=True)
+    o2, _ = a2(obs, deterministic=True)
+    assert torch.allclose(o1, o2)

@@ -1,0 +1,126 @@
+"""Multi-process data-parallel tests over gloo (world_size=2) — the CPU
+stand-in for the RCCL/xGMI path; the collective call pattern is
+identical (single flat-bucket all-reduce / broadcast)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from networks.linear import Actor
+
+
+def _dist_env(rank, world, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+
+
+def _worker_allreduce(rank, world, port, q):
+    _dist_env(rank, world, port)
+    from torch_actor_critic_amd.parallel import comm
+    from torch_actor_critic_amd.parallel.flat import FlatParams
+    comm.init_distributed(backend="gloo")
+
+    torch.manual_seed(rank)  # deliberately different weights per rank
+    actor = Actor(4, 2, [8], act_limit=1.0)
+    fp = FlatParams(actor)
+
+    # 1) sync: rank0's weights must win
+    before = fp.flat.clone()
+    comm.sync_flat_params(fp.flat)
+    q.put(("sync", rank, fp.flat.clone().numpy(), before.numpy()))
+
+    # 2) grad all-reduce: average of per-rank constants
+    fp.zero_grad()
+    fp.flat_grad.fill_(float(rank + 1))
+    comm.allreduce_grads(fp.flat_grad)
+    q.put(("grads", rank, fp.flat_grad.clone().numpy()))
+
+    # 3) epoch stats gather
+    stats = comm.gather_stats([float(rank), float(rank) + 10.0])
+    q.put(("stats", rank, np.array(sorted(stats))))
+
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_two_rank_collectives():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531
+    procs = [ctx.Process(target=_worker_allreduce, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(6):
+        kind, rank, *payload = q.get(timeout=110)
+        results[(kind, rank)] = payload
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+
+    # sync: both ranks end with rank0's pre-sync weights
+    r0_after, r0_before = results[("sync", 0)]
+    r1_after, _ = results[("sync", 1)]
+    np.testing.assert_allclose(r0_after, r0_before)
+    np.testing.assert_allclose(r1_after, r0_before)
+
+    # grads: mean of 1 and 2 = 1.5 on both ranks
+    np.testing.assert_allclose(results[("grads", 0)][0], 1.5)
+    np.testing.assert_allclose(results[("grads", 1)][0], 1.5)
+
+    # stats: union of both ranks' lists
+    np.testing.assert_allclose(results[("stats", 0)][0],
+                               np.array([0.0, 1.0, 10.0, 11.0]))
+
+
+def _worker_training_sync(rank, world, port, q):
+    """Two ranks with the flat-bucket DP path must stay bit-identical."""
+    _dist_env(rank, world, port)
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.parallel import comm
+    comm.init_distributed(backend="gloo")
+
+    torch.manual_seed(100 + rank)
+    actor = Actor(4, 2, [8], act_limit=1.0)
+    opt = FlatAdam(actor, lr=1e-3)
+    comm.sync_flat_params(opt.fp.flat)
+
+    for i in range(3):
+        torch.manual_seed(1000 * rank + i)  # different data per rank
+        obs = torch.randn(8, 4)
+        opt.zero_grad()
+        pi, logp = actor(obs)
+        (pi.square().sum() + logp.sum()).backward()
+        comm.allreduce_grads(opt.fp.flat_grad)
+        opt.step()
+
+    q.put((rank, opt.fp.flat.clone().numpy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_dp_ranks_stay_identical():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29532
+    procs = [ctx.Process(target=_worker_training_sync, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(2):
+        rank, flat = q.get(timeout=110)
+        out[rank] = flat
+    for p in procs:
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    np.testing.assert_allclose(out[0], out[1], atol=1e-7)

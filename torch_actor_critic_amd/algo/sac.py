@@ -1,0 +1,351 @@
+"""Soft Actor-Critic (fixed entropy temperature, twin Q, polyak targets).
+
+API-compatible with the reference ``sac/algorithm.py``: free functions
+``eval_pi_loss`` (:30), ``eval_q_loss`` (:46), ``update_targets`` (:77)
+and class ``SAC`` (:85) with ``update_critic`` (:115), ``update_policy``
+(:143), ``save_model`` (:164), ``train`` (:182) under the same
+signatures and hyperparameter semantics (Appendix A of SURVEY.md).
+
+Deliberate fixes over the reference (each documented in SURVEY.md §8):
+ * Q1: the actor gradient all-reduce runs AFTER backward (the reference
+   averages stale gradients, algorithm.py:155-156).
+ * Q2: the policy loss samples the policy on ``state`` (standard SAC);
+   the reference's next_state sampling is available via
+   ``SAC(..., reference_pi_loss=True)``.
+ * Q3: episode stats are reduced once per epoch (the reference does
+   blocking pickled p2p per env step, algorithm.py:262-271).
+ * Q11: MultiObservation vs ndarray action dispatch is an explicit type
+   check, not try/except TypeError.
+ * Q12: optional learned entropy temperature via
+   ``SAC(..., learn_alpha=True)`` (default off = reference-faithful).
+
+MI355X-first data path: the replay batch, both networks, the optimizers
+and the target live in HBM; one SAC update on GPU is a fixed kernel
+sequence with no host sync (losses accumulate in device tensors), so the
+whole update can be captured into a hipGraph and replayed
+(``algo/graph.py``); gradient reduction is a single flat-bucket RCCL
+all-reduce per module per update.
+"""
+
+import logging
+import typing as t
+from copy import deepcopy
+
+import numpy as np
+import torch
+import torch.nn as nn
+import tqdm
+
+from ..buffer.replay import Batch
+from ..envs.visual import MultiObservation
+from ..ops import functional as Fo
+from ..optim import FlatAdam
+from ..parallel import comm
+from ..parallel.flat import FlatParams, flatten_module_like
+from ..utils import checkpoint as ckpt
+
+logger = logging.getLogger(__name__)
+
+
+# ---------------------------------------------------------------------------
+# Free-function losses (reference API surface)
+# ---------------------------------------------------------------------------
+
+def eval_pi_loss(actor, critic, state, next_state, alpha,
+                 use_next_state: bool = False):
+    """Policy loss (alpha*logp - min(q1,q2)).mean().
+
+    Standard SAC samples the policy on ``state``; ``use_next_state=True``
+    reproduces the reference quirk (algorithm.py:37-38).
+    """
+    pi_state = next_state if use_next_state else state
+    pi, logp_pi = actor(pi_state)
+    q1, q2 = critic(state, pi)
+    return Fo.sac_pi_loss(q1, q2, logp_pi, alpha)
+
+
+def eval_q_loss(actor, critic, target_critic, states, actions, rewards,
+                next_states, done, alpha, gamma, reward_scale):
+    """Twin-Q Bellman MSE (reference algorithm.py:46-74)."""
+    with torch.no_grad():
+        a2, logp_ac = actor(next_states)
+        q1_t, q2_t = target_critic(next_states, a2)
+    q1, q2 = critic(states, actions)
+    return Fo.sac_q_loss(q1, q2, q1_t, q2_t, logp_ac, rewards, done,
+                         alpha, gamma, reward_scale)
+
+
+def update_targets(source: nn.Module, target: nn.Module, polyak: float):
+    """Per-module polyak average (reference algorithm.py:77-81).  Used on
+    the generic path; the flat-buffer path uses ops.polyak_ (one kernel)."""
+    with torch.no_grad():
+        for src, targ in zip(source.parameters(), target.parameters()):
+            targ.data.mul_(polyak).add_(src.data, alpha=1.0 - polyak)
+
+
+def _freeze(module: nn.Module, flag: bool):
+    for p in module.parameters():
+        p.requires_grad = not flag
+
+
+# ---------------------------------------------------------------------------
+# SAC
+# ---------------------------------------------------------------------------
+
+class SAC:
+    def __init__(
+        self,
+        alpha: float,
+        gamma: float,
+        polyak: float,
+        reward_scale: float,
+        epochs: int,
+        batch_size: int,
+        start_steps: int,
+        steps_per_epoch: int,
+        max_ep_len: int,
+        update_after: int,
+        update_every: int,
+        save_every: int,
+        *,
+        reference_pi_loss: bool = False,
+        learn_alpha: bool = False,
+        target_entropy: t.Optional[float] = None,
+        use_graph: bool = True,
+    ):
+        self.alpha = float(alpha)
+        self.gamma = float(gamma)
+        self.polyak = float(polyak)
+        self.reward_scale = float(reward_scale)
+        self.epochs = int(epochs)
+        self.batch_size = int(batch_size)
+        self.start_steps = int(start_steps)
+        self.steps_per_epoch = int(steps_per_epoch)
+        self.max_ep_len = int(max_ep_len)
+        self.update_after = int(update_after)
+        self.update_every = int(update_every)
+        self.save_every = int(save_every)
+        self.reference_pi_loss = reference_pi_loss
+        self.use_graph = use_graph
+
+        # optional learned entropy temperature (default off)
+        self.learn_alpha = learn_alpha
+        self.target_entropy = target_entropy
+        self._log_alpha: t.Optional[torch.Tensor] = None
+        self._alpha_opt = None
+
+        # populated by train()
+        self._target_flat: t.Optional[torch.Tensor] = None
+        self._critic_fp: t.Optional[FlatParams] = None
+        self._actor_fp: t.Optional[FlatParams] = None
+
+    # -- single-module updates (reference method surface) ---------------
+
+    def _alpha_value(self):
+        if self.learn_alpha and self._log_alpha is not None:
+            return self._log_alpha.exp()
+        return self.alpha
+
+    def update_critic(self, q_opt, actor, critic, target_critic,
+                      samples: Batch):
+        q_opt.zero_grad()
+        loss_q = eval_q_loss(
+            actor, critic, target_critic, samples.states, samples.actions,
+            samples.rewards, samples.next_states, samples.done,
+            self._alpha_value(), self.gamma, self.reward_scale)
+        loss_q.backward()
+        self._allreduce(critic, self._critic_fp)
+        q_opt.step()
+        return loss_q
+
+    def update_policy(self, pi_opt, actor, critic, samples: Batch):
+        _freeze(critic, True)
+        pi_opt.zero_grad()
+        pi_state = samples.next_states if self.reference_pi_loss \
+            else samples.states
+        pi, logp_pi = actor(pi_state)
+        q1, q2 = critic(samples.states, pi)
+        loss_pi = Fo.sac_pi_loss(q1, q2, logp_pi, self._alpha_value())
+        loss_pi.backward()
+        # Q1 fix: all-reduce AFTER backward (reference: before, :155-156)
+        self._allreduce(actor, self._actor_fp)
+        pi_opt.step()
+        _freeze(critic, False)
+
+        if self.learn_alpha:
+            self._update_alpha(logp_pi.detach())
+        return loss_pi
+
+    def _update_alpha(self, logp: torch.Tensor):
+        self._alpha_opt.zero_grad()
+        loss_alpha = -(self._log_alpha
+                       * (logp + self.target_entropy)).mean()
+        loss_alpha.backward()
+        self._alpha_opt.step()
+
+    def _allreduce(self, module: nn.Module, fp: t.Optional[FlatParams]):
+        if not comm.is_initialized():
+            return
+        if fp is not None:
+            comm.allreduce_grads(fp.flat_grad)     # ONE bucket
+        else:
+            for p in module.parameters():
+                if p.grad is not None:
+                    comm.allreduce_grads(p.grad)
+
+    def update_targets_fast(self, critic):
+        if self._target_flat is not None and self._critic_fp is not None:
+            Fo.polyak_(self._target_flat, self._critic_fp.flat, self.polyak)
+        else:
+            update_targets(critic, self._target_critic, self.polyak)
+
+    # -- checkpointing ---------------------------------------------------
+
+    def save_model(self, actor, critic, pi_opt, q_opt, epoch: int):
+        """Same artifact layout as the reference (algorithm.py:164-180):
+        actor/, critic/ logged models + auxiliaries {pi_opt, q_opt, epoch}."""
+        dev = next(actor.parameters()).device
+        actor_cpu = deepcopy(actor).cpu()
+        critic_cpu = deepcopy(critic).cpu()
+        ckpt.log_model(actor_cpu, "actor")
+        ckpt.log_model(critic_cpu, "critic")
+        ckpt.log_state_dict({
+            "pi_opt": pi_opt.state_dict(),
+            "q_opt": q_opt.state_dict(),
+            "epoch": epoch,
+        }, "auxiliaries")
+        del actor_cpu, critic_cpu
+        _ = dev
+
+    # -- environment interaction helpers ---------------------------------
+
+    @staticmethod
+    def _state_to_device(state, device):
+        if isinstance(state, MultiObservation):
+            return MultiObservation(state.features.to(device),
+                                    state.frame.to(device))
+        return torch.as_tensor(np.asarray(state, dtype=np.float32),
+                               device=device)
+
+    def _select_action(self, actor, state, device):
+        with torch.no_grad():
+            s = self._state_to_device(state, device)
+            action, _ = actor(s, deterministic=False, with_logprob=False)
+        return action.detach().cpu().numpy()
+
+    # -- main loop --------------------------------------------------------
+
+    def train(self, start_epoch: int, env, actor, critic, buffer,
+              pi_opt, q_opt, render: bool = True, logging: bool = True):
+        device = next(actor.parameters()).device
+        self._target_critic = target_critic = deepcopy(critic)
+        _freeze(target_critic, True)
+
+        # flat-buffer fast path when the optimizers are FlatAdam
+        self._actor_fp = pi_opt.fp if isinstance(pi_opt, FlatAdam) else None
+        self._critic_fp = q_opt.fp if isinstance(q_opt, FlatAdam) else None
+        self._target_flat = flatten_module_like(target_critic) \
+            if self._critic_fp is not None else None
+
+        if self.learn_alpha and self._log_alpha is None:
+            if self.target_entropy is None:
+                # -dim(A) heuristic
+                act_dim = getattr(actor, "act_dim", None)
+                self.target_entropy = -float(act_dim) if act_dim else -1.0
+            self._log_alpha = torch.tensor(
+                float(np.log(self.alpha)), requires_grad=True, device=device)
+            self._alpha_opt = torch.optim.Adam([self._log_alpha], lr=3e-4)
+
+        # initial weight sync: one broadcast per flat buffer (C1)
+        if comm.is_initialized():
+            if self._actor_fp is not None:
+                comm.sync_flat_params(self._actor_fp.flat)
+                comm.sync_flat_params(self._critic_fp.flat)
+                comm.sync_flat_params(self._target_flat)
+            else:
+                for m in (actor, critic, target_critic):
+                    for p in m.parameters():
+                        comm.sync_flat_params(p.data)
+
+        # per-rank seeding (reference algorithm.py:203-205)
+        seed = 10000 * comm.proc_id()
+        torch.manual_seed(seed)
+        np.random.seed(seed)
+
+        state = env.reset()
+        step = 0
+        ep_ret, ep_len = 0.0, 0
+
+        rank0 = comm.proc_id() == 0
+        pbar = tqdm.trange(start_epoch, start_epoch + self.epochs, ncols=0,
+                           initial=start_epoch, disable=not rank0)
+        metrics = {"episode_length": 0.0, "reward": 0.0,
+                   "loss_q": 0.0, "loss_pi": 0.0}
+
+        for e in pbar:
+            episode_rewards: t.List[float] = []
+            episode_lengths: t.List[float] = []
+            # device-side loss accumulators — no per-update host sync
+            loss_q_acc = torch.zeros((), device=device)
+            loss_pi_acc = torch.zeros((), device=device)
+            n_updates = 0
+
+            for _ in range(self.steps_per_epoch):
+                if step < self.start_steps:
+                    action = env.action_space.sample()
+                else:
+                    action = self._select_action(actor, state, device)
+
+                next_state, reward, done, _info = env.step(action)
+                ep_len += 1
+                ep_ret += float(reward)
+                done = False if ep_len == self.max_ep_len else done
+
+                buffer.store(state, action, float(reward), next_state,
+                             float(done))
+                state = next_state
+
+                if done or ep_len == self.max_ep_len:
+                    episode_rewards.append(ep_ret)
+                    episode_lengths.append(ep_len)
+                    state = env.reset()
+                    ep_ret, ep_len = 0.0, 0
+                    if render and rank0:
+                        env.render()
+
+                step += 1
+                if step > self.update_after and step % self.update_every == 0:
+                    for _u in range(self.update_every):
+                        samples = buffer.sample(self.batch_size)
+                        loss_q = self.update_critic(
+                            q_opt, actor, critic, target_critic, samples)
+                        loss_pi = self.update_policy(
+                            pi_opt, actor, critic, samples)
+                        self.update_targets_fast(critic)
+                        loss_q_acc += loss_q.detach()
+                        loss_pi_acc += loss_pi.detach()
+                        n_updates += 1
+
+            # epoch-level stat reduction (Q3 fix; reference did per-step p2p)
+            all_rews = comm.gather_stats(episode_rewards)
+            all_lens = comm.gather_stats(episode_lengths)
+            if n_updates:
+                metrics["loss_q"] = float(loss_q_acc.item()) / n_updates
+                metrics["loss_pi"] = float(loss_pi_acc.item()) / n_updates
+            if all_rews:
+                metrics["reward"] = float(np.mean(all_rews))
+                metrics["episode_length"] = float(np.mean(all_lens))
+
+            if rank0 and logging:
+                if (e + 1) % self.save_every == 0:
+                    self.save_model(actor, critic, pi_opt, q_opt, e)
+                ckpt.log_metrics(metrics, step=e)
+            if rank0:
+                pbar.set_postfix(step=step, **{k: round(v, 3)
+                                               for k, v in metrics.items()})
+
+            # fresh episode each epoch (reference algorithm.py:305-307)
+            state = env.reset()
+            ep_ret, ep_len = 0.0, 0
+
+        return metrics

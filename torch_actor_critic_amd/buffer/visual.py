@@ -1,0 +1,97 @@
+"""Visual replay buffer: dense device arrays, not object arrays.
+
+API-compatible with the reference ``VisualReplayBuffer``
+(``buffer/visual_replay_buffer.py:21-66``: ctor (size, act_dim) —
+vis/feature dims inferred lazily from the first stored observation, since
+the reference stores arbitrary ``MultiObservation`` objects) but stores
+features and frames as two dense tensors per side:
+``features [N, F]`` fp32 and ``frames [N, C, H, W]`` — frames optionally
+uint8-quantized (scale [-1,1] -> u8) to fit 1M 3x64x64 transitions in
+~25 GB of the 288 GB HBM3E instead of ~98 GB fp32.
+"""
+
+import typing as t
+from dataclasses import dataclass
+
+import numpy as np
+import torch
+
+from ..envs.visual import MultiObservation
+
+
+@dataclass(frozen=True)
+class VisualBatch:
+    states: MultiObservation
+    actions: torch.Tensor
+    rewards: torch.Tensor
+    next_states: MultiObservation
+    done: torch.Tensor
+
+
+class VisualReplayBuffer:
+    def __init__(self, size: int, act_dim: int,
+                 device: t.Union[str, torch.device] = "cpu",
+                 seed: int = 0, quantize_frames: bool = True):
+        self.max_size = int(size)
+        self.act_dim = act_dim
+        self.device = torch.device(device)
+        self.quantize = quantize_frames
+        self.ptr = 0
+        self.size = 0
+        self._rng = np.random.default_rng(seed)
+        self._alloc_done = False
+        dev = self.device
+        self.actions = torch.zeros((self.max_size, act_dim),
+                                   dtype=torch.float32, device=dev)
+        self.rewards = torch.zeros(self.max_size, dtype=torch.float32, device=dev)
+        self.done = torch.zeros(self.max_size, dtype=torch.float32, device=dev)
+
+    def _alloc(self, obs: MultiObservation):
+        feat_dim = int(obs.features.numel())
+        vis_dim = tuple(obs.frame.shape)
+        dev = self.device
+        fdt = torch.uint8 if self.quantize else torch.float32
+        self.features = torch.zeros((self.max_size, feat_dim),
+                                    dtype=torch.float32, device=dev)
+        self.frames = torch.zeros((self.max_size, *vis_dim), dtype=fdt, device=dev)
+        self.next_features = torch.zeros_like(self.features)
+        self.next_frames = torch.zeros_like(self.frames)
+        self.feat_dim = feat_dim
+        self.vis_dim = vis_dim
+        self._alloc_done = True
+
+    def _enc_frame(self, frame: torch.Tensor) -> torch.Tensor:
+        if not self.quantize:
+            return frame.to(torch.float32)
+        return ((frame.clamp(-1, 1) + 1.0) * 127.5).round().to(torch.uint8)
+
+    def _dec_frames(self, frames: torch.Tensor) -> torch.Tensor:
+        if not self.quantize:
+            return frames
+        return frames.to(torch.float32) / 127.5 - 1.0
+
+    def store(self, obs: MultiObservation, act, rew,
+              next_obs: MultiObservation, done):
+        if not self._alloc_done:
+            self._alloc(obs)
+        i = self.ptr
+        self.features[i] = obs.features.reshape(-1).to(self.device)
+        self.frames[i] = self._enc_frame(obs.frame.to(self.device))
+        self.next_features[i] = next_obs.features.reshape(-1).to(self.device)
+        self.next_frames[i] = self._enc_frame(next_obs.frame.to(self.device))
+        self.actions[i] = torch.as_tensor(np.asarray(act), dtype=torch.float32
+                                          ).reshape(self.act_dim)
+        self.rewards[i] = float(rew)
+        self.done[i] = float(done)
+        self.ptr = (self.ptr + 1) % self.max_size
+        self.size = min(self.size + 1, self.max_size)
+
+    def sample(self, batch_size: int) -> VisualBatch:
+        idx_np = self._rng.choice(self.size, size=batch_size, replace=False)
+        idx = torch.as_tensor(idx_np, dtype=torch.long, device=self.device)
+        state = MultiObservation(self.features[idx],
+                                 self._dec_frames(self.frames[idx]))
+        next_state = MultiObservation(self.next_features[idx],
+                                      self._dec_frames(self.next_frames[idx]))
+        return VisualBatch(state, self.actions[idx], self.rewards[idx],
+                           next_state, self.done[idx])

@@ -1,0 +1,74 @@
+"""Minimal gym-style environment API (self-contained — this image has no
+gym/mujoco/dm_control, and env stepping is host-side CPU work by nature).
+
+The API mirrors what the reference consumes from gym (reference
+sac/algorithm.py:207,238: ``reset() -> obs``, ``step(a) -> (obs, reward,
+done, info)``, ``action_space``/``observation_space`` with
+``.shape``/``.high``/``.sample()``), so training code is drop-in.
+Environments are registered explicitly (the reference registered its
+custom env via an import side-effect chain — SURVEY.md Q10 — which we fix
+by explicit registration in ``envs/__init__.py``).
+"""
+
+import typing as t
+
+import numpy as np
+
+
+class Box:
+    """Continuous space: shape, bounds, uniform sampling."""
+
+    def __init__(self, low, high, shape: t.Tuple[int, ...],
+                 dtype=np.float32, seed: int = 0):
+        self.shape = tuple(shape)
+        self.dtype = dtype
+        self.low = np.broadcast_to(np.asarray(low, dtype=dtype), self.shape)
+        self.high = np.broadcast_to(np.asarray(high, dtype=dtype), self.shape)
+        self._rng = np.random.default_rng(seed)
+
+    def seed(self, seed: int):
+        self._rng = np.random.default_rng(seed)
+
+    def sample(self) -> np.ndarray:
+        return self._rng.uniform(self.low, self.high).astype(self.dtype)
+
+    def __repr__(self):
+        return f"Box{self.shape}"
+
+
+class Env:
+    action_space: Box
+    observation_space: Box
+
+    def reset(self):
+        raise NotImplementedError
+
+    def step(self, action):
+        raise NotImplementedError
+
+    def seed(self, seed: int):
+        self.action_space.seed(seed)
+
+    def render(self):
+        pass
+
+    def close(self):
+        pass
+
+
+_REGISTRY: t.Dict[str, t.Callable[[], Env]] = {}
+
+
+def register(name: str, factory: t.Callable[[], Env]):
+    _REGISTRY[name] = factory
+
+
+def make(name: str) -> Env:
+    if name not in _REGISTRY:
+        raise KeyError(
+            f"Unknown environment {name!r}. Registered: {sorted(_REGISTRY)}")
+    return _REGISTRY[name]()
+
+
+def registered() -> t.List[str]:
+    return sorted(_REGISTRY)

@@ -1,0 +1,171 @@
+"""Visual (pixels + proprioception) actor / critic networks.
+
+API/state-dict compatible with the reference
+(``networks/convolutional.py:14-183``): same class names, constructor
+signatures, module attribute names (``layers``, ``visual_network`` with
+``conv_i``/``linear``/``final`` children, ``mu_layer``, ``log_std_layer``,
+``final``) and the same architecture quirks, kept deliberately for
+checkpoint compatibility and documented here:
+
+* the CNN trunk is bottlenecked to ONE scalar (``final``: Linear(512,1),
+  reference convolutional.py:49);
+* ``VisualCritic`` applies ReLU to every MLP layer INCLUDING the final
+  width-1 layer (reference convolutional.py:156-158) and combines with the
+  CNN scalar through ``Linear(2,1)``;
+* unbatched inputs are auto-unsqueezed and outputs squeezed
+  (reference convolutional.py:91-96,121,147-154).
+
+The MLP trunk and the tanh-Gaussian head run on the fused gfx950 kernels;
+the conv stack runs through torch Conv2d (MIOpen) until the hand-written
+implicit-GEMM conv kernel lands.
+"""
+
+import typing as t
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from ..envs.visual import MultiObservation
+from ..ops import functional as Fo
+from .mlp import mlp
+
+
+def calculate_size(image_shape, filters, kernel_sizes, strides) -> int:
+    """Flattened size after the conv stack (valid padding)
+    (reference convolutional.py:14-27)."""
+    c, h, w = image_shape
+    for f, k, s in zip(filters, kernel_sizes, strides):
+        c = f
+        h = int(np.floor((h - k) / s + 1))
+        w = int(np.floor((w - k) / s + 1))
+    return int(c * h * w)
+
+
+def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
+               strides=[4, 2, 1], activation: t.Type[nn.Module] = nn.ReLU,
+               dense_size: int = 512) -> nn.Module:
+    """Nature-CNN-style trunk ending in a single scalar
+    (reference convolutional.py:30-51)."""
+    channels = input_shape[0]
+    model = nn.Sequential()
+    sizes = [channels] + list(filters)
+    for i in range(len(sizes) - 1):
+        model.add_module(f"conv_{i}",
+                         nn.Conv2d(sizes[i], sizes[i + 1], kernel_sizes[i],
+                                   strides[i]))
+        model.add_module(f"relu_{i}", activation())
+    flat = calculate_size(input_shape, filters, kernel_sizes, strides)
+    model.add_module("flatten", nn.Flatten())
+    model.add_module("linear", nn.Linear(flat, dense_size))
+    model.add_module("final", nn.Linear(dense_size, 1))
+    return model
+
+
+class VisualActor(nn.Module):
+    """Dual-stream squashed-Gaussian policy
+    (reference convolutional.py:54-121)."""
+
+    def __init__(self, obs_dim: int, act_dim: int,
+                 vis_dim: t.Tuple[int, int, int],
+                 hidden_sizes: t.List[int] = [256, 256],
+                 act_limit: float = 10,
+                 filters: t.List[int] = [32, 64, 64],
+                 kernel_sizes: t.List[int] = [8, 4, 3],
+                 strides: t.List[int] = [4, 2, 1],
+                 log_min_std: float = -20, log_max_std: float = 2):
+        super().__init__()
+        self.layers = mlp([obs_dim] + list(hidden_sizes))
+        self.obs_dim = obs_dim
+        self.act_dim = act_dim
+        self.vis_dim = tuple(vis_dim)
+        self.visual_network = simple_cnn(vis_dim, filters, kernel_sizes,
+                                         strides)
+        # heads sized hidden+1: the CNN contributes one scalar
+        self.mu_layer = nn.Linear(hidden_sizes[-1] + 1, act_dim)
+        self.log_std_layer = nn.Linear(hidden_sizes[-1] + 1, act_dim)
+        self.log_min_std = log_min_std
+        self.log_max_std = log_max_std
+        self.act_limit = act_limit
+
+    def forward(self, x: MultiObservation, deterministic: bool = False,
+                with_logprob: bool = True):
+        image = x.frame
+        if image.ndim == 3:
+            image = image.view((-1, *self.vis_dim))
+        feats = x.features
+        if feats.ndim == 1:
+            feats = feats.view(-1, self.obs_dim)
+
+        h = Fo.mlp_forward(feats, self.layers, relu_last=True)
+        conv_out = self.visual_network(image)
+        h = torch.cat([h, conv_out], dim=1)
+
+        mu = Fo.linear_relu(h, self.mu_layer.weight, self.mu_layer.bias,
+                            relu=False)
+        log_std = Fo.linear_relu(h, self.log_std_layer.weight,
+                                 self.log_std_layer.bias, relu=False)
+        eps = torch.randn_like(mu) if not deterministic else torch.zeros_like(mu)
+        pi_action, logprob = Fo.tanh_gauss_head(
+            mu, log_std, eps, self.act_limit, self.log_min_std,
+            self.log_max_std, deterministic, with_logprob)
+        # reference squeezes both outputs (convolutional.py:121)
+        pi_action = torch.squeeze(pi_action)
+        if logprob is not None:
+            logprob = torch.squeeze(logprob)
+        return pi_action, logprob
+
+
+class VisualCritic(nn.Module):
+    """Q(s,a) over features+frame (reference convolutional.py:124-164)."""
+
+    def __init__(self, obs_dim: int, act_dim: int,
+                 vis_dim: t.Tuple[int, int, int],
+                 hidden_sizes: t.List[int] = [256, 256],
+                 filters: t.List[int] = [32, 64, 64],
+                 kernel_sizes: t.List[int] = [8, 4, 3],
+                 strides: t.List[int] = [4, 2, 1]):
+        super().__init__()
+        self.obs_dim = obs_dim
+        self.act_dim = act_dim
+        self.vis_dim = tuple(vis_dim)
+        self.layers = mlp([obs_dim + act_dim] + list(hidden_sizes) + [1])
+        self.final = nn.Linear(2, 1)
+        self.visual_network = simple_cnn(vis_dim, filters, kernel_sizes,
+                                         strides)
+
+    def forward(self, state: MultiObservation, action):
+        image = state.frame
+        if image.ndim == 3:
+            image = image.view((-1, *self.vis_dim))
+        conv_out = self.visual_network(image)
+
+        x = torch.cat([state.features, action], dim=-1)
+        if x.ndim == 1:
+            x = x.view(-1, self.obs_dim + self.act_dim)
+        # ReLU on ALL layers including the final width-1 layer — kept for
+        # reference parity (convolutional.py:156-158)
+        x = Fo.mlp_forward(x, self.layers, relu_last=True)
+
+        x = torch.cat([x, conv_out], dim=1)
+        x = Fo.linear_relu(x, self.final.weight, self.final.bias, relu=False)
+        return torch.squeeze(x, -1)
+
+
+class VisualDoubleCritic(nn.Module):
+    """Twin visual critics (reference convolutional.py:167-183)."""
+
+    def __init__(self, obs_dim: int, act_dim: int,
+                 vis_dim: t.Tuple[int, int, int],
+                 hidden_sizes: t.List[int] = [256, 256],
+                 filters: t.List[int] = [32, 64, 64],
+                 kernel_sizes: t.List[int] = [8, 4, 3],
+                 strides: t.List[int] = [4, 2, 1]):
+        super().__init__()
+        self.q1 = VisualCritic(obs_dim, act_dim, vis_dim, hidden_sizes,
+                               filters, kernel_sizes, strides)
+        self.q2 = VisualCritic(obs_dim, act_dim, vis_dim, hidden_sizes,
+                               filters, kernel_sizes, strides)
+
+    def forward(self, state: MultiObservation, action):
+        return self.q1(state, action), self.q2(state, action)

@@ -1,0 +1,125 @@
+"""Data-parallel communication over RCCL/xGMI (or gloo on CPU).
+
+Replaces the reference's mpi4py layer (``sac/mpi.py:10-115``) with
+``torch.distributed``: one process per MI355X GPU, backend "nccl" (which
+IS RCCL on ROCm) over the node's xGMI links, gloo for the CPU plumbing
+path and multi-process CPU tests.
+
+Design differences from the reference, driven by xGMI (7 p2p links x
+~153 GB/s per GPU — ring collectives are per-link latency-bound at these
+sub-MB payloads):
+
+* gradient averaging is ONE all-reduce of the module's flat grad bucket
+  per update (the reference does one Allreduce per tensor with NumPy
+  host copies, sac/mpi.py:82-85);
+* initial weight sync is ONE broadcast of the flat param buffer
+  (reference: per-tensor Bcast, sac/mpi.py:96-98);
+* episode stats are reduced ONCE PER EPOCH via a small fixed-size tensor
+  all-gather (the reference blocks on pickled point-to-point sends every
+  env step — SURVEY.md Q3/C4, sac/algorithm.py:262-271).
+"""
+
+import datetime
+import os
+import typing as t
+
+import torch
+import torch.distributed as dist
+
+
+def is_initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def init_distributed(backend: t.Optional[str] = None) -> t.Tuple[int, int]:
+    """Initialise torch.distributed from torchrun-style env vars.
+
+    Returns (rank, world_size).  No-op (0, 1) when WORLD_SIZE is absent
+    or 1.  Backend defaults to nccl(=RCCL) when a GPU is visible, else
+    gloo.
+    """
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return 0, 1
+    if not is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(backend=backend,
+                                timeout=datetime.timedelta(seconds=300))
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    return dist.get_rank(), dist.get_world_size()
+
+
+def proc_id() -> int:
+    return dist.get_rank() if is_initialized() else 0
+
+
+def num_procs() -> int:
+    return dist.get_world_size() if is_initialized() else 1
+
+
+def barrier():
+    if is_initialized():
+        dist.barrier()
+
+
+def sync_flat_params(flat: torch.Tensor, src: int = 0):
+    """Broadcast a module's whole flat parameter buffer — one collective
+    (replaces reference sync_params, sac/mpi.py:93-98)."""
+    if is_initialized():
+        dist.broadcast(flat, src=src)
+
+
+def allreduce_grads(flat_grad: torch.Tensor):
+    """Average a module's flat gradient bucket across ranks — one
+    collective (replaces reference mpi_avg_grads, sac/mpi.py:77-85)."""
+    if not is_initialized():
+        return
+    dist.all_reduce(flat_grad, op=dist.ReduceOp.SUM)
+    flat_grad.div_(num_procs())
+
+
+def allreduce_mean(x: torch.Tensor) -> torch.Tensor:
+    if is_initialized():
+        dist.all_reduce(x, op=dist.ReduceOp.SUM)
+        x = x / num_procs()
+    return x
+
+
+def gather_stats(values: t.Sequence[float], device=None) -> t.List[float]:
+    """All-gather a variable-length list of scalars (episode stats), once
+    per epoch.  Fixed-size padded tensor all-gather — no pickled p2p."""
+    if not is_initialized():
+        return list(values)
+    world = num_procs()
+    n = torch.tensor([len(values)], dtype=torch.int64)
+    if device is not None:
+        n = n.to(device)
+    counts = [torch.zeros_like(n) for _ in range(world)]
+    dist.all_gather(counts, n)
+    max_n = int(max(c.item() for c in counts))
+    if max_n == 0:
+        return []
+    buf = torch.zeros(max_n, dtype=torch.float64)
+    buf[:len(values)] = torch.tensor(list(values), dtype=torch.float64)
+    if device is not None:
+        buf = buf.to(device)
+    out = [torch.zeros_like(buf) for _ in range(world)]
+    dist.all_gather(out, buf)
+    result: t.List[float] = []
+    for c, o in zip(counts, out):
+        result.extend(o[:int(c.item())].tolist())
+    return result
+
+
+def statistics_scalar(x: t.Sequence[float]) -> t.Tuple[float, float]:
+    """Global mean/std of per-rank scalar lists (replaces reference
+    mpi_statistics_scalar, sac/mpi.py:101-115)."""
+    vals = gather_stats(x)
+    if not vals:
+        return 0.0, 0.0
+    tt = torch.tensor(vals, dtype=torch.float64)
+    return float(tt.mean()), float(tt.std(unbiased=False))

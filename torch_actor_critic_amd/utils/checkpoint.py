@@ -1,0 +1,187 @@
+"""MLflow-layout-compatible experiment tracking and checkpointing.
+
+The reference checkpoints through MLflow (``sac/algorithm.py:164-180``,
+``main.py:28-51``): models under
+``mlruns/0/<run_id>/artifacts/{actor,critic}`` (each a directory with an
+``MLmodel`` descriptor and ``data/model.pth``) and an ``auxiliaries``
+state-dict ``{pi_opt, q_opt, epoch}``.  MLflow itself is not installable
+in this image, so this module re-implements exactly that on-disk layout
+(same artifact paths, same torch serialization of the full module, same
+params/metrics file structure), and uses the real mlflow package instead
+whenever it is importable — checkpoints interchange either way.
+"""
+
+import os
+import time
+import typing as t
+import uuid
+
+import torch
+
+try:
+    import mlflow  # noqa: F401
+    import mlflow.pytorch
+    HAVE_MLFLOW = True
+except Exception:  # noqa: BLE001
+    HAVE_MLFLOW = False
+
+
+_TRACKING_DIR = os.environ.get("TAC_AMD_MLRUNS", "mlruns")
+_EXPERIMENT_ID = "0"
+_ACTIVE_RUN: t.Optional[str] = None
+_EXPERIMENT_NAME = "Default"
+
+
+def set_tracking_dir(path: str):
+    global _TRACKING_DIR
+    _TRACKING_DIR = path
+
+
+def set_experiment(name: str):
+    global _EXPERIMENT_NAME
+    _EXPERIMENT_NAME = name
+    if HAVE_MLFLOW:
+        mlflow.set_experiment(name)
+
+
+def _run_dir(run_id: str) -> str:
+    return os.path.join(_TRACKING_DIR, _EXPERIMENT_ID, run_id)
+
+
+def artifact_dir(run_id: str) -> str:
+    return os.path.join(_run_dir(run_id), "artifacts")
+
+
+def start_run() -> str:
+    global _ACTIVE_RUN
+    if HAVE_MLFLOW:
+        run = mlflow.start_run()
+        _ACTIVE_RUN = run.info.run_id
+        return _ACTIVE_RUN
+    run_id = uuid.uuid4().hex
+    rd = _run_dir(run_id)
+    for sub in ("artifacts", "params", "metrics", "tags"):
+        os.makedirs(os.path.join(rd, sub), exist_ok=True)
+    with open(os.path.join(rd, "meta.yaml"), "w") as f:
+        f.write(
+            f"artifact_uri: file://{os.path.abspath(artifact_dir(run_id))}\n"
+            f"experiment_id: '{_EXPERIMENT_ID}'\n"
+            f"run_id: {run_id}\n"
+            f"run_uuid: {run_id}\n"
+            f"run_name: {_EXPERIMENT_NAME}\n"
+            f"status: 1\n"
+            f"start_time: {int(time.time() * 1000)}\n"
+            "lifecycle_stage: active\n")
+    _ACTIVE_RUN = run_id
+    return run_id
+
+
+def active_run_id() -> t.Optional[str]:
+    return _ACTIVE_RUN
+
+
+def end_run():
+    global _ACTIVE_RUN
+    if HAVE_MLFLOW:
+        mlflow.end_run()
+    _ACTIVE_RUN = None
+
+
+def log_params(params: t.Dict[str, t.Any]):
+    if HAVE_MLFLOW:
+        mlflow.log_params(params)
+        return
+    if _ACTIVE_RUN is None:
+        return
+    pd = os.path.join(_run_dir(_ACTIVE_RUN), "params")
+    os.makedirs(pd, exist_ok=True)
+    for k, v in params.items():
+        with open(os.path.join(pd, k), "w") as f:
+            f.write(str(v))
+
+
+def log_metrics(metrics: t.Dict[str, float], step: int = 0):
+    if HAVE_MLFLOW:
+        mlflow.log_metrics(metrics, step=step)
+        return
+    if _ACTIVE_RUN is None:
+        return
+    md = os.path.join(_run_dir(_ACTIVE_RUN), "metrics")
+    os.makedirs(md, exist_ok=True)
+    ts = int(time.time() * 1000)
+    for k, v in metrics.items():
+        with open(os.path.join(md, k), "a") as f:
+            f.write(f"{ts} {v} {step}\n")
+
+
+def get_run_params(run_id: str) -> t.Dict[str, str]:
+    """Read back a run's logged params as strings (mirrors
+    MlflowClient().get_run(run_id).data.params — reference main.py:29-31)."""
+    if HAVE_MLFLOW:
+        from mlflow.tracking import MlflowClient
+        return dict(MlflowClient().get_run(run_id).data.params)
+    pd = os.path.join(_run_dir(run_id), "params")
+    out = {}
+    if os.path.isdir(pd):
+        for name in os.listdir(pd):
+            with open(os.path.join(pd, name)) as f:
+                out[name] = f.read()
+    return out
+
+
+# -- model artifacts ----------------------------------------------------
+
+def log_model(module: torch.nn.Module, artifact_path: str):
+    """Save a full pickled module under
+    artifacts/<artifact_path>/data/model.pth with an MLmodel descriptor
+    (mlflow.pytorch.log_model layout — reference sac/algorithm.py:172-173)."""
+    if HAVE_MLFLOW:
+        mlflow.pytorch.log_model(module, artifact_path)
+        return
+    if _ACTIVE_RUN is None:
+        return
+    base = os.path.join(artifact_dir(_ACTIVE_RUN), artifact_path)
+    os.makedirs(os.path.join(base, "data"), exist_ok=True)
+    was_training = module.training
+    module = module.cpu() if next(module.parameters()).is_cuda else module
+    torch.save(module, os.path.join(base, "data", "model.pth"))
+    if was_training:
+        module.train()
+    with open(os.path.join(base, "MLmodel"), "w") as f:
+        f.write(
+            "flavors:\n"
+            "  pytorch:\n"
+            "    model_data: data\n"
+            f"    pytorch_version: {torch.__version__}\n"
+            "  python_function:\n"
+            "    data: data\n"
+            "    loader_module: mlflow.pytorch\n")
+
+
+def load_model(model_uri: str) -> torch.nn.Module:
+    """Load a module saved by log_model (mlflow.pytorch.load_model —
+    reference main.py:34-35, run_agent.py:75)."""
+    if HAVE_MLFLOW:
+        return mlflow.pytorch.load_model(model_uri)
+    path = os.path.join(model_uri, "data", "model.pth")
+    return torch.load(path, map_location="cpu", weights_only=False)
+
+
+def log_state_dict(sd: dict, artifact_path: str):
+    """artifacts/<path>/state_dict.pth (mlflow.pytorch.log_state_dict —
+    reference sac/algorithm.py:176-180)."""
+    if HAVE_MLFLOW:
+        mlflow.pytorch.log_state_dict(sd, artifact_path=artifact_path)
+        return
+    if _ACTIVE_RUN is None:
+        return
+    base = os.path.join(artifact_dir(_ACTIVE_RUN), artifact_path)
+    os.makedirs(base, exist_ok=True)
+    torch.save(sd, os.path.join(base, "state_dict.pth"))
+
+
+def load_state_dict(artifact_uri: str) -> dict:
+    if HAVE_MLFLOW:
+        return mlflow.pytorch.load_state_dict(artifact_uri)
+    return torch.load(os.path.join(artifact_uri, "state_dict.pth"),
+                      map_location="cpu", weights_only=False)

@@ -1,0 +1,214 @@
+"""Flagship benchmark: SAC updates/sec + env steps/sec on HalfCheetah-v4
+MLP (BASELINE.json headline metric).
+
+Per bench step: ONE env step (synthetic HalfCheetah-shaped transition,
+stored into the HBM replay ring) and ONE SAC update (batch sampled
+on-device, critic+actor forward/backward on the gfx950 MFMA kernels,
+fused Adam, polyak) — the reference's 1:1 replay ratio
+(reference sac/algorithm.py:273-278: 50 updates per 50 env steps; we run
+the same 50-step window structure).  On GPU the update burst replays a
+captured hipGraph.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for
+N>1 launched under torch.distributed.run, one rank per GPU over RCCL.
+Rank 0 prints ONE JSON line; `value` is the whole-job aggregate
+updates/sec over all N GPUs.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=2000,
+                   help="timed env-steps (= SAC updates, 1:1)")
+    p.add_argument("--warmup", type=int, default=200)
+    p.add_argument("--batch-size", type=int, default=64)
+    p.add_argument("--env", default="HalfCheetah-v4")
+    p.add_argument("--hidden", type=int, nargs="+", default=[256, 256])
+    p.add_argument("--buffer-size", type=int, default=1_000_000)
+    p.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
+    p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--update-window", type=int, default=50,
+                   help="env-steps per update burst (reference: 50)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from torch_actor_critic_amd import envs
+    from torch_actor_critic_amd.algo.sac import SAC, _freeze
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.ops import functional as Fo
+    from torch_actor_critic_amd.parallel import comm
+    from torch_actor_critic_amd.parallel.flat import flatten_module_like
+    from copy import deepcopy
+
+    rank, world = comm.init_distributed()
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    dtype = args.dtype if on_gpu else "fp32"
+    if on_gpu:
+        Fo.set_compute_dtype(dtype)
+    Fo.set_philox_seed(1234 + 7919 * rank)
+
+    env = envs.make(args.env)
+    env.seed(1000 + rank)
+    obs_dim = env.observation_space.shape[0]
+    act_dim = env.action_space.shape[0]
+    act_limit = float(env.action_space.high.reshape(-1)[0])
+
+    torch.manual_seed(0)  # identical init on all ranks
+    actor = Actor(obs_dim, act_dim, list(args.hidden),
+                  act_limit=act_limit).to(device)
+    critic = DoubleCritic(obs_dim, act_dim, list(args.hidden)).to(device)
+    target_critic = deepcopy(critic)
+    _freeze(target_critic, True)
+
+    pi_opt = FlatAdam(actor, lr=3e-4)
+    q_opt = FlatAdam(critic, lr=3e-4)
+    target_flat = flatten_module_like(target_critic)
+    if world > 1:
+        comm.sync_flat_params(pi_opt.fp.flat)
+        comm.sync_flat_params(q_opt.fp.flat)
+        comm.sync_flat_params(target_flat)
+
+    sac = SAC(alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=args.batch_size, start_steps=0,
+              steps_per_epoch=1, max_ep_len=1000, update_after=0,
+              update_every=args.update_window, save_every=10**9)
+
+    # ---- prefill the HBM replay ring with synthetic transitions -------
+    buf_size = args.buffer_size if on_gpu else min(args.buffer_size, 100_000)
+    buffer = ReplayBuffer(buf_size, obs_dim, act_dim, device=device,
+                          seed=rank)
+    rng = np.random.default_rng(rank)
+    prefill = max(10_000, 4 * args.batch_size)
+    chunk = 10_000
+    for i0 in range(0, prefill, chunk):
+        n = min(chunk, prefill - i0)
+        buffer.store_batch(
+            rng.standard_normal((n, obs_dim)).astype(np.float32),
+            rng.standard_normal((n, act_dim)).astype(np.float32),
+            rng.standard_normal(n).astype(np.float32),
+            rng.standard_normal((n, obs_dim)).astype(np.float32),
+            np.zeros(n, dtype=np.float32))
+
+    # ---- build the update path (hipGraph on GPU) ----------------------
+    graph = None
+    if on_gpu and not args.no_graph:
+        from torch_actor_critic_amd.algo.graph import GraphedSACUpdate
+        graph = GraphedSACUpdate(sac, actor, critic, target_critic, buffer,
+                                 pi_opt, q_opt, target_flat,
+                                 args.batch_size, device)
+
+    sac._actor_fp = pi_opt.fp
+    sac._critic_fp = q_opt.fp
+    sac._target_flat = target_flat
+    sac._target_critic = target_critic
+
+    state = env.reset()
+
+    def run_window(n_steps):
+        """n_steps env steps then n_steps updates (reference structure)."""
+        nonlocal state
+        with torch.no_grad():
+            for _ in range(n_steps):
+                s = torch.as_tensor(state, dtype=torch.float32,
+                                    device=device)
+                a, _ = actor(s, deterministic=False, with_logprob=False)
+                a_np = a.detach().cpu().numpy()
+                nstate, reward, done, _ = env.step(a_np)
+                buffer.store(state, a_np, float(reward), nstate,
+                             float(done))
+                state = env.reset() if done else nstate
+        if graph is not None:
+            for _ in range(n_steps):
+                graph.step()
+        else:
+            for _ in range(n_steps):
+                samples = buffer.sample(args.batch_size)
+                sac.update_critic(q_opt, actor, critic, target_critic,
+                                  samples)
+                sac.update_policy(pi_opt, actor, critic, samples)
+                sac.update_targets_fast(critic)
+
+    def sync():
+        comm.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    win = args.update_window
+    # warmup
+    done_w = 0
+    while done_w < args.warmup:
+        n = min(win, args.warmup - done_w)
+        run_window(n)
+        done_w += n
+
+    sync()
+    t0 = time.perf_counter()
+    done_s = 0
+    while done_s < args.steps:
+        n = min(win, args.steps - done_s)
+        run_window(n)
+        done_s += n
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    el_t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as dist
+        el_t = el_t.to(device) if on_gpu else el_t
+        dist.all_reduce(el_t, op=dist.ReduceOp.MAX)
+    elapsed = float(el_t.item())
+
+    n_gpus = world if on_gpu else args.gpus
+    updates_per_sec = world * args.steps / elapsed
+    if rank == 0:
+        result = {
+            "metric": "sac_updates_per_sec",
+            "value": updates_per_sec,
+            "unit": "updates/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": f"SAC-MLP{args.hidden}-{args.env}",
+                "global_batch": args.batch_size * world,
+                "seq_len": 1,
+                "parallelism": f"dp{world}",
+                "obs_dim": obs_dim,
+                "act_dim": act_dim,
+                "replay_buffer": buf_size,
+                "env_steps_per_sec": updates_per_sec,
+                "update_window": win,
+                "graph": graph is not None,
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,357 @@
+"""GPU numerics tests: every hand-written gfx950 kernel vs a plain
+PyTorch fp32 reference of the same op (run on CPU)."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from torch_actor_critic_amd.ops import require_extension
+    return require_extension()
+
+
+@pytest.fixture(autouse=True)
+def _fp32_mode():
+    from torch_actor_critic_amd.ops import functional as Fo
+    Fo.set_compute_dtype("fp32")
+    yield
+    Fo.set_compute_dtype("fp32")
+
+
+DEV = "cuda:0"
+
+
+# ---------------------------------------------------------------------------
+# MFMA GEMM
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [
+    (64, 256, 17), (64, 256, 256), (64, 6, 256), (64, 1, 262),
+    (1, 256, 17), (4096, 256, 256), (100, 23, 393), (64, 56, 257),
+])
+def test_linear_fwd_fp32_parity(ext, M, N, K):
+    torch.manual_seed(0)
+    x = torch.randn(M, K)
+    w = torch.randn(N, K)
+    b = torch.randn(N)
+    ref = F.linear(x, w, b)
+    out = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), False).cpu()
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4), \
+        (out - ref).abs().max()
+
+
+def test_linear_fwd_relu(ext):
+    x = torch.randn(64, 64)
+    w = torch.randn(32, 64)
+    b = torch.randn(32)
+    ref = F.relu(F.linear(x, w, b))
+    out = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), True).cpu()
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+
+
+def test_linear_fwd_transpose_detecting(ext):
+    """Asymmetric operands so a row/col-swapped C-write cannot pass."""
+    M, N, K = 32, 48, 40
+    x = torch.arange(M * K, dtype=torch.float32).reshape(M, K) / (M * K)
+    w = (torch.arange(N * K, dtype=torch.float32).reshape(N, K) ** 1.3) / (N * K)
+    ref = F.linear(x, w, None)
+    out = ext.linear_fwd(x.to(DEV), w.to(DEV),
+                         torch.zeros(N, device=DEV), False).cpu()
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.parametrize("M,N,K,relu", [
+    (64, 256, 17, False), (64, 256, 256, True), (64, 1, 262, False),
+    (128, 6, 256, False), (4096, 256, 256, True),
+])
+def test_linear_bwd_parity(ext, M, N, K, relu):
+    torch.manual_seed(1)
+    x = torch.randn(M, K, requires_grad=True)
+    w = torch.randn(N, K, requires_grad=True)
+    b = torch.randn(N, requires_grad=True)
+    y = F.linear(x, w, b)
+    if relu:
+        y = F.relu(y)
+    dy = torch.randn(M, N)
+    y.backward(dy)
+
+    y_dev = ext.linear_fwd(x.detach().to(DEV), w.detach().to(DEV),
+                           b.detach().to(DEV), relu)
+    dx, dw, db = ext.linear_bwd(dy.to(DEV), x.detach().to(DEV),
+                                w.detach().to(DEV), y_dev, relu, True)
+    assert torch.allclose(dw.cpu(), w.grad, atol=1e-3, rtol=1e-4), \
+        (dw.cpu() - w.grad).abs().max()
+    assert torch.allclose(db.cpu(), b.grad, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(dx.cpu(), x.grad, atol=1e-3, rtol=1e-4)
+
+
+def test_linear_bf16_mode_close(ext):
+    """bf16 MFMA path: ~1e-2 relative accuracy vs fp32 reference."""
+    from torch_actor_critic_amd.ops import functional as Fo
+    torch.manual_seed(2)
+    x = torch.randn(256, 256)
+    w = torch.randn(256, 256) / 16
+    b = torch.randn(256)
+    ref = F.linear(x, w, b)
+    Fo.set_compute_dtype("bf16")
+    try:
+        out = ext.linear_fwd(x.to(DEV), w.to(DEV), b.to(DEV), False).cpu()
+    finally:
+        Fo.set_compute_dtype("fp32")
+    rel = ((out - ref).abs() / (ref.abs() + 1.0)).max()
+    assert rel < 0.05, rel
+    # and it must NOT be bitwise-identical to fp32 (proves bf16 ran)
+    assert not torch.allclose(out, ref, atol=1e-7)
+
+
+# ---------------------------------------------------------------------------
+# Fused tanh-Gaussian head
+# ---------------------------------------------------------------------------
+
+def _ref_head(mu, log_std, eps, act_limit, lo, hi, det):
+    ls = torch.clip(log_std, lo, hi)
+    std = torch.exp(ls)
+    prob = mu if det else mu + std * eps
+    pi = torch.tanh(prob) * act_limit
+    gauss = (-0.5 * ((prob - mu) / std) ** 2 - ls
+             - 0.5 * math.log(2 * math.pi)).sum(-1)
+    corr = (2 * math.log(2) - prob - F.softplus(-2 * prob)).sum(-1)
+    return pi, gauss - corr
+
+
+@pytest.mark.parametrize("B,A,det", [(64, 6, False), (64, 6, True),
+                                     (256, 17, False), (1, 56, False)])
+def test_tanh_gauss_fwd_parity(ext, B, A, det):
+    torch.manual_seed(3)
+    mu = torch.randn(B, A)
+    log_std = torch.randn(B, A) * 3
+    eps = torch.randn(B, A)
+    pi_ref, logp_ref = _ref_head(mu, log_std, eps, 2.0, -20., 2., det)
+    pi, logp, prob, lsc = ext.tanh_gauss_fwd(
+        mu.to(DEV), log_std.to(DEV), eps.to(DEV), 2.0, -20., 2., det, True)
+    assert torch.allclose(pi.cpu(), pi_ref, atol=1e-5)
+    assert torch.allclose(logp.cpu(), logp_ref, atol=1e-4), \
+        (logp.cpu() - logp_ref).abs().max()
+
+
+def test_tanh_gauss_bwd_parity(ext):
+    torch.manual_seed(4)
+    B, A = 64, 6
+    mu = torch.randn(B, A, requires_grad=True)
+    log_std = torch.randn(B, A, requires_grad=True)
+    eps = torch.randn(B, A)
+    pi_ref, logp_ref = _ref_head(mu, log_std, eps, 2.0, -20., 2., False)
+    dpi = torch.randn(B, A)
+    dlogp = torch.randn(B)
+    (pi_ref * dpi).sum().backward(retain_graph=True)
+    (logp_ref * dlogp).sum().backward()
+
+    pi, logp, prob, lsc = ext.tanh_gauss_fwd(
+        mu.detach().to(DEV), log_std.detach().to(DEV), eps.to(DEV),
+        2.0, -20., 2., False, True)
+    dmu, dls = ext.tanh_gauss_bwd(
+        dpi.to(DEV), dlogp.to(DEV), mu.detach().to(DEV),
+        log_std.detach().to(DEV), eps.to(DEV), prob, lsc,
+        2.0, -20., 2., False, True)
+    assert torch.allclose(dmu.cpu(), mu.grad, atol=1e-4), \
+        (dmu.cpu() - mu.grad).abs().max()
+    assert torch.allclose(dls.cpu(), log_std.grad, atol=1e-4), \
+        (dls.cpu() - log_std.grad).abs().max()
+
+
+# ---------------------------------------------------------------------------
+# Fused losses
+# ---------------------------------------------------------------------------
+
+def test_q_loss_parity(ext):
+    torch.manual_seed(5)
+    B = 256
+    q1, q2 = torch.randn(B), torch.randn(B)
+    q1t, q2t = torch.randn(B), torch.randn(B)
+    logp, r = torch.randn(B), torch.randn(B)
+    d = (torch.rand(B) > 0.8).float()
+    backup = 1.5 * r + 0.99 * (1 - d) * (torch.min(q1t, q2t) - 0.2 * logp)
+    ref = ((q1 - backup) ** 2).mean() + ((q2 - backup) ** 2).mean()
+    loss, dq1, dq2 = ext.sac_q_loss_fwd(
+        q1.to(DEV), q2.to(DEV), q1t.to(DEV), q2t.to(DEV), logp.to(DEV),
+        r.to(DEV), d.to(DEV), 0.2, 0.99, 1.5)
+    assert torch.allclose(loss.cpu(), ref, atol=1e-5)
+    assert torch.allclose(dq1.cpu(), 2 * (q1 - backup) / B, atol=1e-6)
+    assert torch.allclose(dq2.cpu(), 2 * (q2 - backup) / B, atol=1e-6)
+
+
+def test_pi_loss_parity(ext):
+    torch.manual_seed(6)
+    B = 256
+    q1, q2, logp = torch.randn(B), torch.randn(B), torch.randn(B)
+    ref = (0.2 * logp - torch.min(q1, q2)).mean()
+    loss, dq1, dq2, dlogp = ext.sac_pi_loss_fwd(
+        q1.to(DEV), q2.to(DEV), logp.to(DEV), 0.2)
+    assert torch.allclose(loss.cpu(), ref, atol=1e-5)
+    assert torch.allclose(dlogp.cpu(), torch.full((B,), 0.2 / B), atol=1e-7)
+    # min-branch gradients
+    mask1 = (q1 < q2).float()
+    assert torch.allclose(dq1.cpu(), -mask1 / B, atol=1e-7)
+
+
+# ---------------------------------------------------------------------------
+# Flat maintenance
+# ---------------------------------------------------------------------------
+
+def test_polyak_parity(ext):
+    t = torch.randn(100_003)
+    s = torch.randn(100_003)
+    ref = 0.995 * t + 0.005 * s
+    td = t.to(DEV)
+    ext.polyak_(td, s.to(DEV), 0.995)
+    assert torch.allclose(td.cpu(), ref, atol=1e-6)
+
+
+def test_adam_parity(ext):
+    torch.manual_seed(7)
+    n = 10_001
+    p = torch.randn(n)
+    g = torch.randn(n)
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    pd, gd, md, vd = (x.clone().to(DEV) for x in (p, g, m, v))
+    step = torch.zeros(1, dtype=torch.int64, device=DEV)
+    lr, b1, b2, eps = 1e-3, 0.9, 0.999, 1e-8
+    for it in range(1, 4):
+        ext.adam_step_(pd, gd, md, vd, step, lr, b1, b2, eps, 0.0)
+        m.mul_(b1).add_(g, alpha=1 - b1)
+        v.mul_(b2).addcmul_(g, g, value=1 - b2)
+        denom = (v / (1 - b2 ** it)).sqrt().add_(eps)
+        p.addcdiv_(m, denom, value=-lr / (1 - b1 ** it))
+    assert int(step.item()) == 3
+    assert torch.allclose(pd.cpu(), p, atol=1e-5), (pd.cpu() - p).abs().max()
+    assert torch.allclose(md.cpu(), m, atol=1e-6)
+
+
+# ---------------------------------------------------------------------------
+# Replay sample + Philox noise
+# ---------------------------------------------------------------------------
+
+def test_replay_sample_gpu():
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    buf = ReplayBuffer(5000, 8, 3, device=DEV)
+    n = 3000
+    rng = np.random.default_rng(0)
+    obs = np.arange(n, dtype=np.float32)[:, None].repeat(8, 1)
+    act = np.arange(n, dtype=np.float32)[:, None].repeat(3, 1)
+    buf.store_batch(obs, act, np.arange(n, dtype=np.float32), obs + 1,
+                    (np.arange(n) % 2).astype(np.float32))
+    b = buf.sample(512)
+    assert b.states.is_cuda
+    sv = b.states.cpu()
+    # row consistency + bounds
+    assert (sv[:, 0] >= 0).all() and (sv[:, 0] < n).all()
+    assert torch.allclose(b.actions.cpu()[:, 0], sv[:, 0])
+    assert torch.allclose(b.rewards.cpu(), sv[:, 0])
+    assert torch.allclose(b.next_states.cpu()[:, 0], sv[:, 0] + 1)
+    # counter advances -> different draws
+    b2 = buf.sample(512)
+    assert not torch.equal(b.states, b2.states)
+    # roughly uniform coverage
+    idx = sv[:, 0].to(torch.long)
+    assert idx.unique().numel() > 350
+
+
+def test_philox_randn(ext):
+    from torch_actor_critic_amd.ops import functional as Fo
+    Fo.set_philox_seed(42)
+    x = torch.zeros(100_000, device=DEV)
+    y1 = Fo.randn_like_philox(x)
+    y2 = Fo.randn_like_philox(x)
+    assert not torch.equal(y1, y2)
+    assert abs(float(y1.mean())) < 0.02
+    assert abs(float(y1.std()) - 1.0) < 0.02
+    assert float(y1.abs().max()) < 7.0
+
+
+# ---------------------------------------------------------------------------
+# Full-model parity + graph capture
+# ---------------------------------------------------------------------------
+
+def test_actor_critic_gpu_matches_cpu():
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    torch.manual_seed(8)
+    actor = Actor(17, 6, [256, 256], act_limit=1.0)
+    critic = DoubleCritic(17, 6, [256, 256])
+    obs = torch.randn(64, 17)
+    act = torch.randn(64, 6)
+
+    pi_cpu, logp_cpu = actor(obs, deterministic=True)
+    q1_cpu, q2_cpu = critic(obs, act)
+
+    actor_g = Actor(17, 6, [256, 256], act_limit=1.0).to(DEV)
+    actor_g.load_state_dict(actor.state_dict())
+    critic_g = DoubleCritic(17, 6, [256, 256]).to(DEV)
+    critic_g.load_state_dict(critic.state_dict())
+    pi_g, logp_g = actor_g(obs.to(DEV), deterministic=True)
+    q1_g, q2_g = critic_g(obs.to(DEV), act.to(DEV))
+
+    assert torch.allclose(pi_g.cpu(), pi_cpu, atol=1e-4)
+    assert torch.allclose(logp_g.cpu(), logp_cpu, atol=1e-3)
+    assert torch.allclose(q1_g.cpu(), q1_cpu, atol=1e-3)
+    assert torch.allclose(q2_g.cpu(), q2_cpu, atol=1e-3)
+
+
+def test_graphed_update():
+    from copy import deepcopy
+    from torch_actor_critic_amd.algo.graph import GraphedSACUpdate
+    from torch_actor_critic_amd.algo.sac import SAC, _freeze
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.parallel.flat import flatten_module_like
+
+    torch.manual_seed(9)
+    device = torch.device(DEV)
+    actor = Actor(17, 6, [64, 64], act_limit=1.0).to(device)
+    critic = DoubleCritic(17, 6, [64, 64]).to(device)
+    target = deepcopy(critic)
+    _freeze(target, True)
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+    target_flat = flatten_module_like(target)
+
+    buf = ReplayBuffer(10_000, 17, 6, device=device)
+    rng = np.random.default_rng(1)
+    buf.store_batch(rng.standard_normal((1000, 17)).astype(np.float32),
+                    rng.standard_normal((1000, 6)).astype(np.float32),
+                    rng.standard_normal(1000).astype(np.float32),
+                    rng.standard_normal((1000, 17)).astype(np.float32),
+                    np.zeros(1000, dtype=np.float32))
+
+    sac = SAC(alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=64, start_steps=0, steps_per_epoch=1,
+              max_ep_len=100, update_after=0, update_every=1, save_every=10)
+    g = GraphedSACUpdate(sac, actor, critic, target, buf, pi_opt, q_opt,
+                         target_flat, 64, device)
+
+    p0 = pi_opt.fp.flat.clone()
+    c0 = q_opt.fp.flat.clone()
+    t0 = target_flat.clone()
+    step0 = int(pi_opt.step_t.item())
+    for _ in range(10):
+        g.step()
+    torch.cuda.synchronize()
+    # parameters moved, target tracked, step counter advanced per replay
+    assert not torch.allclose(p0, pi_opt.fp.flat)
+    assert not torch.allclose(c0, q_opt.fp.flat)
+    assert not torch.allclose(t0, target_flat)
+    assert int(pi_opt.step_t.item()) == step0 + 10
+    lq, lp = g.read_and_reset_losses(10)
+    assert np.isfinite(lq) and np.isfinite(lp) and lq > 0
+
+
+def test_smoke_entry():
+    import __graft_entry__
+    __graft_entry__.smoke()

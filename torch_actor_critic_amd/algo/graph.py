@@ -1,0 +1,157 @@
+"""hipGraph-captured SAC update.
+
+One SAC update at batch 64 is ~60 tiny kernels; eager-launched that is
+~60 x 3.5 us of host launch latency — the dominant cost on MI355X (the
+whole MLP forward is microseconds of GPU work).  This module captures the
+ENTIRE update step — replay sample+gather, critic forward/backward, fused
+Adam, policy forward/backward, fused Adam, polyak — into a hipGraph
+(``torch.cuda.CUDAGraph`` == hipGraph on ROCm) and replays it per update:
+one ~10-16 us replay instead of ~200 us of launches.
+
+Graph-replay safety is designed in everywhere:
+ * replay sampling / policy noise / Adam step counters live in DEVICE
+   memory and are bumped by tiny same-stream predecessor kernels
+   (ops/csrc/tac_kernels.hip), so each replay draws fresh indices/noise
+   and correct bias corrections;
+ * all parameters/gradients are stable flat buffers (parallel/flat.py);
+ * losses accumulate into static device tensors — zero host sync.
+
+Data-parallel mode (world>1) splits the capture into three graphs with
+the two flat-bucket RCCL all-reduces between them (collective capture is
+backend-dependent; the split costs two eager launches per update).
+"""
+
+import logging
+import typing as t
+
+import torch
+
+from ..buffer.replay import Batch
+from ..optim import FlatAdam
+from ..parallel import comm
+from . import sac as sac_mod
+
+logger = logging.getLogger(__name__)
+
+
+class GraphedSACUpdate:
+    def __init__(self, sac, actor, critic, target_critic, buffer,
+                 pi_opt: FlatAdam, q_opt: FlatAdam,
+                 target_flat: torch.Tensor, batch_size: int,
+                 device: torch.device, warmup_iters: int = 3):
+        self.sac = sac
+        self.actor = actor
+        self.critic = critic
+        self.target_critic = target_critic
+        self.buffer = buffer
+        self.pi_opt = pi_opt
+        self.q_opt = q_opt
+        self.target_flat = target_flat
+        self.device = device
+        self.world = comm.num_procs()
+
+        obs_dim = buffer.obs_dim
+        act_dim = buffer.act_dim
+        opts = dict(device=device, dtype=torch.float32)
+        self.batch = Batch(
+            states=torch.zeros(batch_size, obs_dim, **opts),
+            actions=torch.zeros(batch_size, act_dim, **opts),
+            rewards=torch.zeros(batch_size, **opts),
+            next_states=torch.zeros(batch_size, obs_dim, **opts),
+            done=torch.zeros(batch_size, **opts),
+        )
+        # static loss accumulators (read at epoch boundaries only)
+        self.loss_q_acc = torch.zeros((), **opts)
+        self.loss_pi_acc = torch.zeros((), **opts)
+
+        # -- warmup on a side stream (per torch.cuda.graph contract) ----
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup_iters):
+                self._phase_critic()
+                self._reduce(self.q_opt)
+                self._phase_policy()
+                self._reduce(self.pi_opt)
+                self._phase_finish()
+        torch.cuda.current_stream().wait_stream(s)
+
+        # -- capture ----------------------------------------------------
+        if self.world == 1:
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self._phase_critic()
+                self._phase_policy()
+                self._phase_finish()
+            self._graphs = None
+        else:
+            g1 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g1):
+                self._phase_critic()
+            g2 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g2, pool=g1.pool()):
+                self._phase_policy()
+            g3 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g3, pool=g1.pool()):
+                self._phase_finish()
+            self.graph = None
+            self._graphs = (g1, g2, g3)
+
+    # -- phases (recorded into the graph) -------------------------------
+
+    def _phase_critic(self):
+        self.buffer.sample_into(self.batch)
+        self.q_opt.zero_grad()
+        loss_q = sac_mod.eval_q_loss(
+            self.actor, self.critic, self.target_critic,
+            self.batch.states, self.batch.actions, self.batch.rewards,
+            self.batch.next_states, self.batch.done,
+            self.sac.alpha, self.sac.gamma, self.sac.reward_scale)
+        loss_q.backward()
+        self.loss_q_acc += loss_q.detach()
+
+    def _phase_policy(self):
+        self.q_opt.step()
+        sac_mod._freeze(self.critic, True)
+        self.pi_opt.zero_grad()
+        pi_state = self.batch.next_states if self.sac.reference_pi_loss \
+            else self.batch.states
+        pi, logp = self.actor(pi_state)
+        q1, q2 = self.critic(self.batch.states, pi)
+        from ..ops import functional as Fo
+        loss_pi = Fo.sac_pi_loss(q1, q2, logp, self.sac.alpha)
+        loss_pi.backward()
+        sac_mod._freeze(self.critic, False)
+        self.loss_pi_acc += loss_pi.detach()
+
+    def _phase_finish(self):
+        self.pi_opt.step()
+        from ..ops import functional as Fo
+        Fo.polyak_(self.target_flat, self.q_opt.fp.flat, self.sac.polyak)
+
+    def _reduce(self, opt: FlatAdam):
+        if self.world > 1:
+            comm.allreduce_grads(opt.fp.flat_grad)
+
+    # -- replay ----------------------------------------------------------
+
+    def step(self):
+        if self.graph is not None:
+            self.graph.replay()
+        else:
+            g1, g2, g3 = self._graphs
+            g1.replay()
+            self._reduce(self.q_opt)
+            g2.replay()
+            self._reduce(self.pi_opt)
+            g3.replay()
+
+    def read_and_reset_losses(self, n_updates: int) -> t.Tuple[float, float]:
+        """Host-syncs ONCE (epoch boundary): mean losses since last call."""
+        if n_updates <= 0:
+            return 0.0, 0.0
+        lq = float(self.loss_q_acc.item()) / n_updates
+        lp = float(self.loss_pi_acc.item()) / n_updates
+        self.loss_q_acc.zero_()
+        self.loss_pi_acc.zero_()
+        return lq, lp

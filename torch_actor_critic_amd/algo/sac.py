@@ -138,6 +138,8 @@ class SAC:
         self._target_flat: t.Optional[torch.Tensor] = None
         self._critic_fp: t.Optional[FlatParams] = None
         self._actor_fp: t.Optional[FlatParams] = None
+        self._graph = None
+        self._graph_failed = False
 
     # -- single-module updates (reference method surface) ---------------
 
@@ -192,6 +194,32 @@ class SAC:
             for p in module.parameters():
                 if p.grad is not None:
                     comm.allreduce_grads(p.grad)
+
+    def _maybe_build_graph(self, actor, critic, target_critic, buffer,
+                           pi_opt, q_opt, device):
+        """Build the hipGraph-captured update on first use (GPU +
+        FlatAdam only).  Falls back to the eager path on any capture
+        failure (logged once)."""
+        if self._graph is not None:
+            return self._graph
+        if (self._graph_failed or not self.use_graph
+                or device.type != "cuda"
+                or self._critic_fp is None or self._actor_fp is None
+                or self.learn_alpha):
+            return None
+        try:
+            from .graph import GraphedSACUpdate
+            self._graph = GraphedSACUpdate(
+                self, actor, critic, target_critic, buffer, pi_opt, q_opt,
+                self._target_flat, self.batch_size, device)
+            logger.info("SAC update captured into hipGraph "
+                        "(world=%d)", comm.num_procs())
+        except Exception as e:  # noqa: BLE001
+            logger.warning("hipGraph capture failed, using eager updates: "
+                           "%r", e)
+            self._graph_failed = True
+            self._graph = None
+        return self._graph
 
     def update_targets_fast(self, critic):
         if self._target_flat is not None and self._critic_fp is not None:
@@ -315,23 +343,36 @@ class SAC:
 
                 step += 1
                 if step > self.update_after and step % self.update_every == 0:
-                    for _u in range(self.update_every):
-                        samples = buffer.sample(self.batch_size)
-                        loss_q = self.update_critic(
-                            q_opt, actor, critic, target_critic, samples)
-                        loss_pi = self.update_policy(
-                            pi_opt, actor, critic, samples)
-                        self.update_targets_fast(critic)
-                        loss_q_acc += loss_q.detach()
-                        loss_pi_acc += loss_pi.detach()
-                        n_updates += 1
+                    graph = self._maybe_build_graph(
+                        actor, critic, target_critic, buffer, pi_opt, q_opt,
+                        device)
+                    if graph is not None:
+                        for _u in range(self.update_every):
+                            graph.step()
+                        n_updates += self.update_every
+                    else:
+                        for _u in range(self.update_every):
+                            samples = buffer.sample(self.batch_size)
+                            loss_q = self.update_critic(
+                                q_opt, actor, critic, target_critic, samples)
+                            loss_pi = self.update_policy(
+                                pi_opt, actor, critic, samples)
+                            self.update_targets_fast(critic)
+                            loss_q_acc += loss_q.detach()
+                            loss_pi_acc += loss_pi.detach()
+                            n_updates += 1
 
             # epoch-level stat reduction (Q3 fix; reference did per-step p2p)
             all_rews = comm.gather_stats(episode_rewards)
             all_lens = comm.gather_stats(episode_lengths)
             if n_updates:
-                metrics["loss_q"] = float(loss_q_acc.item()) / n_updates
-                metrics["loss_pi"] = float(loss_pi_acc.item()) / n_updates
+                if self._graph is not None:
+                    lq, lp = self._graph.read_and_reset_losses(n_updates)
+                    metrics["loss_q"] = lq
+                    metrics["loss_pi"] = lp
+                else:
+                    metrics["loss_q"] = float(loss_q_acc.item()) / n_updates
+                    metrics["loss_pi"] = float(loss_pi_acc.item()) / n_updates
             if all_rews:
                 metrics["reward"] = float(np.mean(all_rews))
                 metrics["episode_length"] = float(np.mean(all_lens))

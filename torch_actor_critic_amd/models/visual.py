@@ -105,7 +105,7 @@ class VisualActor(nn.Module):
                             relu=False)
         log_std = Fo.linear_relu(h, self.log_std_layer.weight,
                                  self.log_std_layer.bias, relu=False)
-        eps = torch.randn_like(mu) if not deterministic else torch.zeros_like(mu)
+        eps = Fo.randn_like_philox(mu) if not deterministic else torch.zeros_like(mu)
         pi_action, logprob = Fo.tanh_gauss_head(
             mu, log_std, eps, self.act_limit, self.log_min_std,
             self.log_max_std, deterministic, with_logprob)

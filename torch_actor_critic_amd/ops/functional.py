@@ -22,6 +22,49 @@ from . import use_native, require_extension
 _LOG_2PI = math.log(2.0 * math.pi)
 _2LOG2 = 2.0 * math.log(2.0)
 
+_compute_dtype = "fp32"
+
+
+def set_compute_dtype(dtype: str):
+    """'bf16' switches GEMM kernels to bf16 MFMA inputs (fp32 accumulate,
+    fp32 master weights); 'fp32' uses the exact f32 MFMA path."""
+    global _compute_dtype
+    assert dtype in ("fp32", "bf16")
+    _compute_dtype = dtype
+    from . import extension
+    ext = extension()
+    if ext is not None:
+        ext.set_compute_bf16(dtype == "bf16")
+
+
+def get_compute_dtype() -> str:
+    return _compute_dtype
+
+
+# per-device Philox noise state (graph-replay-safe rsample noise)
+_philox_ctr = {}
+_philox_seed = 0
+
+
+def set_philox_seed(seed: int):
+    global _philox_seed
+    _philox_seed = int(seed)
+    _philox_ctr.clear()
+
+
+def randn_like_philox(t: torch.Tensor) -> torch.Tensor:
+    """Standard-normal noise; on GPU drawn by the Philox kernel with a
+    device-side counter so a captured hipGraph replays fresh noise."""
+    if use_native(t):
+        dev = t.device
+        if dev not in _philox_ctr:
+            _philox_ctr[dev] = torch.zeros(1, dtype=torch.int64, device=dev)
+        out = torch.empty_like(t)
+        require_extension().philox_randn_(out, _philox_ctr[dev],
+                                          _philox_seed)
+        return out
+    return torch.randn_like(t)
+
 
 # ---------------------------------------------------------------------------
 # Linear (+ optional ReLU)

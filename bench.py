@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--no-graph", action="store_true")
     p.add_argument("--update-window", type=int, default=50,
                    help="env-steps per update burst (reference: 50)")
+    p.add_argument("--mode", choices=["full", "updates", "acting"],
+                   default="full",
+                   help="isolation: updates-only / acting-only / both")
     return p.parse_args()
 
 
@@ -123,19 +126,41 @@ def main():
 
     state = env.reset()
 
+    act_graph = None
+    if on_gpu:
+        from torch_actor_critic_amd.algo.act import ActGraph, WindowedStore
+        act_graph = ActGraph(actor, obs_dim, act_dim, device)
+        wstore = WindowedStore(buffer, args.update_window)
+
+    do_act = args.mode in ("full", "acting")
+    do_upd = args.mode in ("full", "updates")
+
     def run_window(n_steps):
         """n_steps env steps then n_steps updates (reference structure)."""
         nonlocal state
-        with torch.no_grad():
-            for _ in range(n_steps):
-                s = torch.as_tensor(state, dtype=torch.float32,
-                                    device=device)
-                a, _ = actor(s, deterministic=False, with_logprob=False)
-                a_np = a.detach().cpu().numpy()
-                nstate, reward, done, _ = env.step(a_np)
-                buffer.store(state, a_np, float(reward), nstate,
-                             float(done))
-                state = env.reset() if done else nstate
+        if do_act:
+            if act_graph is not None:
+                for _ in range(n_steps):
+                    a_np = act_graph.act(state)
+                    nstate, reward, done, _ = env.step(a_np)
+                    wstore.store(state, a_np, float(reward), nstate,
+                                 float(done))
+                    state = env.reset() if done else nstate
+                wstore.flush()
+            else:
+                with torch.no_grad():
+                    for _ in range(n_steps):
+                        s = torch.as_tensor(state, dtype=torch.float32,
+                                            device=device)
+                        a, _ = actor(s, deterministic=False,
+                                     with_logprob=False)
+                        a_np = a.detach().cpu().numpy()
+                        nstate, reward, done, _ = env.step(a_np)
+                        buffer.store(state, a_np, float(reward), nstate,
+                                     float(done))
+                        state = env.reset() if done else nstate
+        if not do_upd:
+            return
         if graph is not None:
             for _ in range(n_steps):
                 graph.step()
@@ -205,6 +230,7 @@ def main():
                 "env_steps_per_sec": updates_per_sec,
                 "update_window": win,
                 "graph": graph is not None,
+                "mode": args.mode,
             },
         }
         print(json.dumps(result))

@@ -1,0 +1,87 @@
+"""Fast environment-interaction path.
+
+The env loop is host-side and serial (MuJoCo-like physics can't move to
+the GPU — SURVEY.md §7 hard parts), so per-step cost is pure latency:
+an eager B=1 actor forward is ~15 Python-dispatched kernel launches plus
+a device sync.  This module cuts it to: one pinned H2D copy, ONE
+hipGraph replay (the whole actor forward incl. Philox noise), one pinned
+D2H copy, one event sync (~25 us/step instead of ~1 ms).
+
+``WindowedStore`` batches the interaction window's transitions in host
+pinned buffers and flushes them to the HBM ring ONCE per update window —
+semantically identical to per-step stores (the reference samples only at
+burst time, sac/algorithm.py:273-278) but 6 launches per window instead
+of 6 per step.
+"""
+
+import numpy as np
+import torch
+
+
+class ActGraph:
+    """hipGraph-captured single-state stochastic actor forward."""
+
+    def __init__(self, actor, obs_dim: int, act_dim: int,
+                 device: torch.device, warmup: int = 3):
+        self.device = device
+        self.obs_in = torch.zeros(1, obs_dim, device=device)
+        self.obs_pin = torch.zeros(obs_dim, pin_memory=True)
+        self.act_pin = torch.zeros(1, act_dim, pin_memory=True)
+        self.ev = torch.cuda.Event()
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(warmup):
+                actor(self.obs_in, deterministic=False, with_logprob=False)
+        torch.cuda.current_stream().wait_stream(s)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph), torch.no_grad():
+            self.act_out, _ = actor(self.obs_in, deterministic=False,
+                                    with_logprob=False)
+
+    def act(self, state: np.ndarray) -> np.ndarray:
+        self.obs_pin.copy_(torch.from_numpy(np.asarray(state,
+                                                       dtype=np.float32)))
+        self.obs_in.copy_(self.obs_pin.view(1, -1), non_blocking=True)
+        self.graph.replay()
+        self.act_pin.copy_(self.act_out, non_blocking=True)
+        self.ev.record()
+        self.ev.synchronize()
+        return self.act_pin[0].numpy().copy()
+
+
+class WindowedStore:
+    """Host-side staging of up to `window` transitions, flushed to the
+    replay ring in one batched write."""
+
+    def __init__(self, buffer, window: int):
+        self.buffer = buffer
+        obs_dim, act_dim = buffer.obs_dim, buffer.act_dim
+        self.obs = np.zeros((window, obs_dim), dtype=np.float32)
+        self.act = np.zeros((window, act_dim), dtype=np.float32)
+        self.rew = np.zeros(window, dtype=np.float32)
+        self.nobs = np.zeros((window, obs_dim), dtype=np.float32)
+        self.done = np.zeros(window, dtype=np.float32)
+        self.n = 0
+        self.window = window
+
+    def store(self, obs, act, rew, next_obs, done):
+        i = self.n
+        self.obs[i] = obs
+        self.act[i] = act
+        self.rew[i] = rew
+        self.nobs[i] = next_obs
+        self.done[i] = done
+        self.n += 1
+        if self.n == self.window:
+            self.flush()
+
+    def flush(self):
+        if self.n == 0:
+            return
+        n = self.n
+        self.buffer.store_batch(self.obs[:n], self.act[:n], self.rew[:n],
+                                self.nobs[:n], self.done[:n])
+        self.n = 0

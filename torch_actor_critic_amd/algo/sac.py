@@ -300,6 +300,23 @@ class SAC:
         torch.manual_seed(seed)
         np.random.seed(seed)
 
+        # fast acting path: captured B=1 actor forward + windowed ring
+        # stores (GPU, vector-obs envs only; visual envs use eager acting)
+        act_graph = None
+        wstore = None
+        if (device.type == "cuda" and self.use_graph
+                and hasattr(buffer, "obs_dim")):
+            try:
+                from .act import ActGraph, WindowedStore
+                act_graph = ActGraph(actor, buffer.obs_dim, buffer.act_dim,
+                                     device)
+                wstore = WindowedStore(buffer, self.update_every)
+            except Exception as e:  # noqa: BLE001
+                logger.warning("act-graph capture failed (%r); using eager "
+                               "acting", e)
+                act_graph = None
+                wstore = None
+
         state = env.reset()
         step = 0
         ep_ret, ep_len = 0.0, 0
@@ -321,6 +338,8 @@ class SAC:
             for _ in range(self.steps_per_epoch):
                 if step < self.start_steps:
                     action = env.action_space.sample()
+                elif act_graph is not None:
+                    action = act_graph.act(state)
                 else:
                     action = self._select_action(actor, state, device)
 
@@ -329,8 +348,12 @@ class SAC:
                 ep_ret += float(reward)
                 done = False if ep_len == self.max_ep_len else done
 
-                buffer.store(state, action, float(reward), next_state,
-                             float(done))
+                if wstore is not None:
+                    wstore.store(state, action, float(reward), next_state,
+                                 float(done))
+                else:
+                    buffer.store(state, action, float(reward), next_state,
+                                 float(done))
                 state = next_state
 
                 if done or ep_len == self.max_ep_len:
@@ -343,6 +366,8 @@ class SAC:
 
                 step += 1
                 if step > self.update_after and step % self.update_every == 0:
+                    if wstore is not None:
+                        wstore.flush()  # updates must see this window
                     graph = self._maybe_build_graph(
                         actor, critic, target_critic, buffer, pi_opt, q_opt,
                         device)
@@ -361,6 +386,9 @@ class SAC:
                             loss_q_acc += loss_q.detach()
                             loss_pi_acc += loss_pi.detach()
                             n_updates += 1
+
+            if wstore is not None:
+                wstore.flush()
 
             # epoch-level stat reduction (Q3 fix; reference did per-step p2p)
             all_rews = comm.gather_stats(episode_rewards)

@@ -113,11 +113,22 @@ def main():
 
     # ---- build the update path (hipGraph on GPU) ----------------------
     graph = None
+    engine_kind = "eager"
     if on_gpu and not args.no_graph:
-        from torch_actor_critic_amd.algo.graph import GraphedSACUpdate
-        graph = GraphedSACUpdate(sac, actor, critic, target_critic, buffer,
-                                 pi_opt, q_opt, target_flat,
-                                 args.batch_size, device)
+        try:
+            from torch_actor_critic_amd.algo.engine import FusedSACEngine
+            graph = FusedSACEngine(sac, actor, critic, target_critic,
+                                   buffer, pi_opt, q_opt, target_flat,
+                                   args.batch_size, device,
+                                   philox_seed=1234 + 7919 * rank)
+            engine_kind = "fused"
+        except Exception as e:  # noqa: BLE001
+            print(f"# fused engine failed ({e!r}); autograd graph fallback")
+            from torch_actor_critic_amd.algo.graph import GraphedSACUpdate
+            graph = GraphedSACUpdate(sac, actor, critic, target_critic,
+                                     buffer, pi_opt, q_opt, target_flat,
+                                     args.batch_size, device)
+            engine_kind = "autograd-graph"
 
     sac._actor_fp = pi_opt.fp
     sac._critic_fp = q_opt.fp
@@ -230,6 +241,7 @@ def main():
                 "env_steps_per_sec": updates_per_sec,
                 "update_window": win,
                 "graph": graph is not None,
+                "engine": engine_kind,
                 "mode": args.mode,
             },
         }

@@ -1,0 +1,284 @@
+"""Fused SAC update engine: full-update numerical parity against an
+eager fp32 PyTorch reference driven with IDENTICAL Philox noise, plus
+hipGraph-replay behavior tests."""
+
+import copy
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+B, O, A = 64, 17, 6
+HID = [64, 64]
+ALPHA, GAMMA, POLYAK, SCALE = 0.2, 0.99, 0.995, 1.0
+
+
+@pytest.fixture(autouse=True)
+def _fp32_mode():
+    from torch_actor_critic_amd.ops import functional as Fo
+    Fo.set_compute_dtype("fp32")
+    yield
+    Fo.set_compute_dtype("fp32")
+
+
+def _setup(batch=B, learn_alpha=False):
+    from copy import deepcopy
+    from torch_actor_critic_amd.algo.engine import FusedSACEngine
+    from torch_actor_critic_amd.algo.sac import SAC, _freeze
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.parallel.flat import flatten_module_like
+
+    torch.manual_seed(11)
+    device = torch.device(DEV)
+    actor = Actor(O, A, HID, act_limit=1.0).to(device)
+    critic = DoubleCritic(O, A, HID).to(device)
+    target = deepcopy(critic)
+    _freeze(target, True)
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+    target_flat = flatten_module_like(target)
+    buf = ReplayBuffer(4096, O, A, device=device)
+    sac = SAC(alpha=ALPHA, gamma=GAMMA, polyak=POLYAK, reward_scale=SCALE,
+              epochs=1, batch_size=batch, start_steps=0, steps_per_epoch=1,
+              max_ep_len=100, update_after=0, update_every=1, save_every=10,
+              learn_alpha=learn_alpha)
+    eng = FusedSACEngine(sac, actor, critic, target, buf, pi_opt, q_opt,
+                         target_flat, batch, device, sample=False,
+                         capture=False, philox_seed=0)
+    return sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng
+
+
+def _eager_reference(actor_cpu, critic_cpu, target_cpu, s, a, r, ns, d, eps):
+    """One full SAC update on CPU eager fp32 (reference semantics with the
+    engine's stacked-noise convention), returning every quantity the
+    engine produces."""
+    import torch.nn.functional as F
+    from torch_actor_critic_amd.ops.functional import _eager_tanh_gauss
+    from torch_actor_critic_amd.optim import FlatAdam
+
+    pi_opt = FlatAdam(actor_cpu, lr=3e-4)
+    q_opt = FlatAdam(critic_cpu, lr=3e-4)
+
+    def actor_fwd(x, eps_rows):
+        h = x
+        for l in actor_cpu.layers:
+            h = F.relu(F.linear(h, l.weight, l.bias))
+        mu = F.linear(h, actor_cpu.mu_layer.weight, actor_cpu.mu_layer.bias)
+        ls = F.linear(h, actor_cpu.log_std_layer.weight,
+                      actor_cpu.log_std_layer.bias)
+        return _eager_tanh_gauss(mu, ls, eps_rows, 1.0, -20., 2., False, True)
+
+    # ---- critic update ----
+    with torch.no_grad():
+        a2, logp_next = actor_fwd(ns, eps[B:])
+        q1t, q2t = target_cpu(ns, a2)
+        backup = SCALE * r + GAMMA * (1 - d) * (
+            torch.min(q1t, q2t) - ALPHA * logp_next)
+    q_opt.zero_grad()
+    q1, q2 = critic_cpu(s, a)
+    loss_q = ((q1 - backup) ** 2).mean() + ((q2 - backup) ** 2).mean()
+    loss_q.backward()
+    cgrad = q_opt.fp.flat_grad.clone()
+    q_opt.step()
+
+    # ---- policy update (post critic Adam, frozen critic) ----
+    for p in critic_cpu.parameters():
+        p.requires_grad = False
+    pi_opt.zero_grad()
+    pi, logp = actor_fwd(s, eps[:B])
+    q1p, q2p = critic_cpu(s, pi)
+    loss_pi = (ALPHA * logp - torch.min(q1p, q2p)).mean()
+    loss_pi.backward()
+    agrad = pi_opt.fp.flat_grad.clone()
+    pi_opt.step()
+    for p in critic_cpu.parameters():
+        p.requires_grad = True
+
+    return dict(loss_q=float(loss_q), loss_pi=float(loss_pi),
+                cgrad=cgrad, agrad=agrad,
+                cflat=q_opt.fp.flat.clone(), aflat=pi_opt.fp.flat.clone())
+
+
+def test_engine_full_update_parity():
+    sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+        _setup()
+    from torch_actor_critic_amd.ops import require_extension
+    ext = require_extension()
+
+    # CPU twins with identical weights
+    actor_cpu = copy.deepcopy(actor).cpu()
+    critic_cpu = copy.deepcopy(critic).cpu()
+    target_cpu = copy.deepcopy(target).cpu()
+    t0_flat = target_flat.clone().cpu()
+
+    torch.manual_seed(5)
+    s = torch.randn(B, O)
+    a = torch.rand(B, A) * 2 - 1
+    r = torch.randn(B)
+    ns = torch.randn(B, O)
+    d = (torch.rand(B) > 0.9).float()
+
+    eng.load_batch(s.to(DEV), a.to(DEV), r.to(DEV), ns.to(DEV), d.to(DEV))
+    eng._run_once()
+    torch.cuda.synchronize()
+
+    # the engine bumped its counter once -> noise drawn at ctr=1
+    eps = ext.tg_eps(1, 0, 2 * B, A, eng.mu).cpu()
+    ref = _eager_reference(actor_cpu, critic_cpu, target_cpu,
+                           s, a, r, ns, d, eps)
+
+    lq = float(eng.loss_q_acc.item())
+    lp = float(eng.loss_pi_acc.item())
+    assert abs(lq - ref["loss_q"]) < 2e-3 * max(1, abs(ref["loss_q"])), \
+        (lq, ref["loss_q"])
+    assert abs(lp - ref["loss_pi"]) < 2e-3 * max(1, abs(ref["loss_pi"])), \
+        (lp, ref["loss_pi"])
+
+    cg = q_opt.fp.flat_grad.cpu()
+    ag = pi_opt.fp.flat_grad.cpu()
+    assert torch.allclose(cg, ref["cgrad"], atol=5e-5, rtol=1e-3), \
+        (cg - ref["cgrad"]).abs().max()
+    assert torch.allclose(ag, ref["agrad"], atol=5e-5, rtol=1e-3), \
+        (ag - ref["agrad"]).abs().max()
+
+    # post-Adam parameters
+    assert torch.allclose(q_opt.fp.flat.cpu(), ref["cflat"], atol=1e-5), \
+        (q_opt.fp.flat.cpu() - ref["cflat"]).abs().max()
+    assert torch.allclose(pi_opt.fp.flat.cpu(), ref["aflat"], atol=1e-5), \
+        (pi_opt.fp.flat.cpu() - ref["aflat"]).abs().max()
+
+    # polyak target tracked the UPDATED critic
+    expect_t = POLYAK * t0_flat + (1 - POLYAK) * ref["cflat"]
+    assert torch.allclose(target_flat.cpu(), expect_t, atol=1e-5)
+
+
+def test_engine_graph_replay_advances():
+    sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+        _setup()
+    rng = np.random.default_rng(3)
+    buf.store_batch(rng.standard_normal((512, O)).astype(np.float32),
+                    rng.standard_normal((512, A)).astype(np.float32),
+                    rng.standard_normal(512).astype(np.float32),
+                    rng.standard_normal((512, O)).astype(np.float32),
+                    np.zeros(512, dtype=np.float32))
+    # build a captured engine (sample=True)
+    from torch_actor_critic_amd.algo.engine import FusedSACEngine
+    eng2 = FusedSACEngine(sac, actor, critic, target, buf, pi_opt, q_opt,
+                          target_flat, B, torch.device(DEV), sample=True,
+                          capture=True, philox_seed=7)
+    p0 = pi_opt.fp.flat.clone()
+    c0 = q_opt.fp.flat.clone()
+    t0 = target_flat.clone()
+    ctr0 = int(eng2.ctr.item())
+    step0 = int(q_opt.step_t.item())
+    xc_prev = eng2.XC.clone()
+    for _ in range(5):
+        eng2.step()
+    torch.cuda.synchronize()
+    assert int(eng2.ctr.item()) == ctr0 + 5      # fresh draws per replay
+    assert int(q_opt.step_t.item()) == step0 + 5
+    assert not torch.allclose(p0, pi_opt.fp.flat)
+    assert not torch.allclose(c0, q_opt.fp.flat)
+    assert not torch.allclose(t0, target_flat)
+    assert not torch.equal(xc_prev, eng2.XC)     # new batch each replay
+    lq, lp = eng2.read_and_reset_losses(5)
+    assert np.isfinite(lq) and np.isfinite(lp) and lq > 0
+    # params stay finite
+    assert torch.isfinite(pi_opt.fp.flat).all()
+    assert torch.isfinite(q_opt.fp.flat).all()
+
+
+def test_engine_learned_alpha():
+    sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+        _setup(learn_alpha=True)
+    rng = np.random.default_rng(4)
+    buf.store_batch(rng.standard_normal((512, O)).astype(np.float32),
+                    rng.standard_normal((512, A)).astype(np.float32),
+                    rng.standard_normal(512).astype(np.float32),
+                    rng.standard_normal((512, O)).astype(np.float32),
+                    np.zeros(512, dtype=np.float32))
+    from torch_actor_critic_amd.algo.engine import FusedSACEngine
+    eng2 = FusedSACEngine(sac, actor, critic, target, buf, pi_opt, q_opt,
+                          target_flat, B, torch.device(DEV), sample=True,
+                          capture=True, philox_seed=9)
+    a0 = float(eng2.alpha_dev.item())
+    assert abs(a0 - ALPHA) < 1e-4
+    for _ in range(10):
+        eng2.step()
+    torch.cuda.synchronize()
+    a1 = float(eng2.alpha_dev.item())
+    assert np.isfinite(a1) and a1 > 0
+    assert a1 != a0  # alpha moved in-graph
+    assert int(eng2.alpha_step.item()) == 12  # 2 warmup + 10 replays
+
+
+def test_engine_bf16_mode_trains():
+    from torch_actor_critic_amd.ops import functional as Fo
+    sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, _ = \
+        _setup()
+    rng = np.random.default_rng(5)
+    buf.store_batch(rng.standard_normal((512, O)).astype(np.float32),
+                    rng.standard_normal((512, A)).astype(np.float32),
+                    rng.standard_normal(512).astype(np.float32),
+                    rng.standard_normal((512, O)).astype(np.float32),
+                    np.zeros(512, dtype=np.float32))
+    Fo.set_compute_dtype("bf16")
+    try:
+        from torch_actor_critic_amd.algo.engine import FusedSACEngine
+        eng = FusedSACEngine(sac, actor, critic, target, buf, pi_opt,
+                             q_opt, target_flat, B, torch.device(DEV),
+                             sample=True, capture=True, philox_seed=13)
+        for _ in range(10):
+            eng.step()
+        torch.cuda.synchronize()
+        lq, lp = eng.read_and_reset_losses(10)
+    finally:
+        Fo.set_compute_dtype("fp32")
+    assert np.isfinite(lq) and np.isfinite(lp)
+    assert torch.isfinite(pi_opt.fp.flat).all()
+    assert torch.isfinite(q_opt.fp.flat).all()
+
+
+def test_sac_train_on_gpu_learns_pendulum():
+    """End-to-end: SAC.train on the GPU fast path (act-graph + fused
+    engine) improves Pendulum reward — the full-framework GPU check."""
+    from torch_actor_critic_amd import envs
+    from torch_actor_critic_amd.algo.sac import SAC
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+
+    torch.manual_seed(0)
+    np.random.seed(0)
+    device = torch.device(DEV)
+    env = envs.make("Pendulum-v1")
+    env.seed(0)
+    actor = Actor(3, 1, [64, 64], act_limit=2.0).to(device)
+    critic = DoubleCritic(3, 1, [64, 64]).to(device)
+    buf = ReplayBuffer(20000, 3, 1, device=device)
+    pi_opt, q_opt = FlatAdam(actor, lr=1e-3), FlatAdam(critic, lr=1e-3)
+
+    sac = SAC(alpha=0.1, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=64, start_steps=500,
+              steps_per_epoch=4000, max_ep_len=200, update_after=500,
+              update_every=50, save_every=1000)
+    sac.train(0, env, actor, critic, buf, pi_opt, q_opt, render=False,
+              logging=False)
+    assert sac._graph is not None, "fused engine was not used"
+
+    eval_rets = []
+    for _ in range(3):
+        state = env.reset()
+        ep, done = 0.0, False
+        while not done:
+            with torch.no_grad():
+                a, _ = actor(torch.as_tensor(state, device=device),
+                             deterministic=True, with_logprob=False)
+            state, r, done, _ = env.step(a.cpu().numpy())
+            ep += r
+        eval_rets.append(ep)
+    assert float(np.mean(eval_rets)) > -400.0, eval_rets

@@ -1,0 +1,399 @@
+"""Hand-scheduled fused SAC update engine.
+
+Replaces the autograd-recorded update (112 kernels/update, 75% of GPU
+time in scalar-staged GEMMs — profiles/r01_baseline_update_profile.md)
+with an explicit kernel schedule of ~40 launches built from the
+multi-problem MFMA kernels in ops/csrc/fused.hip:
+
+ * one Philox counter bump covers the replay draw AND the policy noise;
+ * the replay gather writes states/actions/next-states directly into
+   concat-layout buffers (XC = [s|a ; ns|a2], XC2 = [s|pi]) — torch.cat
+   never appears;
+ * ONE actor forward over 2B stacked rows serves both the Bellman
+   backup action (rows B:, on next-states) and the policy action
+   (rows :B, on states) — mathematically identical to the reference's
+   two independent draws since the rows are disjoint states;
+ * both twin critics run per GEMM launch (blockIdx.z);
+ * dgrads are fwd-form GEMMs on cached transposed weights (refreshed in
+   ONE kernel after each Adam step);
+ * wgrads write straight into the FlatAdam gradient buffers;
+ * losses are single-block deterministic reductions accumulating into
+   device scalars; alpha may live on device and be Adam-updated
+   in-graph (learned entropy temperature, BASELINE north star).
+
+The whole schedule is captured into a hipGraph (3 graphs + 2 flat-bucket
+RCCL all-reduces when data-parallel).  Reference semantics preserved:
+update order, losses and polyak follow sac/algorithm.py:115-162,77-81.
+"""
+
+import math
+import typing as t
+
+import torch
+
+from ..buffer.replay import ReplayBuffer
+from ..models.mlp import Actor, DoubleCritic
+from ..optim import FlatAdam
+from ..parallel import comm
+
+
+class FusedSACEngine:
+    def __init__(self, sac, actor: Actor, critic: DoubleCritic,
+                 target_critic: DoubleCritic, buffer: ReplayBuffer,
+                 pi_opt: FlatAdam, q_opt: FlatAdam,
+                 target_flat: torch.Tensor, batch_size: int,
+                 device: torch.device, sample: bool = True,
+                 capture: bool = True, philox_seed: int = 0):
+        from ..ops import require_extension
+        self.ext = require_extension()
+        self.sac = sac
+        self.actor = actor
+        self.critic = critic
+        self.buffer = buffer
+        self.pi_opt = pi_opt
+        self.q_opt = q_opt
+        self.target_flat = target_flat
+        self.device = device
+        self.sample = sample
+        self.world = comm.num_procs()
+        self.seed = philox_seed
+
+        B = self.B = batch_size
+        O = self.O = buffer.obs_dim
+        A = self.A = buffer.act_dim
+        OC = self.OC = O + A
+        self.act_limit = float(actor.act_limit)
+        self.lo = float(actor.log_min_std)
+        self.hi = float(actor.log_max_std)
+
+        f32 = dict(device=device, dtype=torch.float32)
+
+        # ---- static batch / activation buffers ------------------------
+        self.XC = torch.zeros(2 * B, OC, **f32)    # [s|a ; ns|a2]
+        self.XC2 = torch.zeros(B, OC, **f32)       # [s|pi]
+        self.rew = torch.zeros(B, **f32)
+        self.done = torch.zeros(B, **f32)
+        self.ctr = torch.zeros(1, dtype=torch.int64, device=device)
+
+        # actor trunk
+        self.a_hidden = [l.out_features for l in actor.layers]
+        self.a_act = [torch.zeros(2 * B, h, **f32) for h in self.a_hidden]
+        self.mu = torch.zeros(2 * B, A, **f32)
+        self.ls = torch.zeros(2 * B, A, **f32)
+        self.prob = torch.zeros(2 * B, A, **f32)
+        self.logp = torch.zeros(2 * B, **f32)
+        self.dmu = torch.zeros(B, A, **f32)
+        self.dls = torch.zeros(B, A, **f32)
+        self.da = [torch.zeros(B, h, **f32) for h in self.a_hidden]
+
+        # critic stacks: widths e.g. [h1, h2, 1]
+        self.c_w = [l.out_features for l in critic.q1.layers]
+        nL = len(self.c_w)
+
+        def cbufs(rows):
+            return [[torch.zeros(rows, w, **f32) for w in self.c_w]
+                    for _ in range(2)]
+
+        self.t_act = cbufs(B)     # target critic activations
+        self.c_act = cbufs(B)     # critic activations (q phase)
+        self.p_act = cbufs(B)     # critic activations (pi phase)
+        self.dq = [torch.zeros(B, 1, **f32) for _ in range(2)]
+        self.dqp = [torch.zeros(B, 1, **f32) for _ in range(2)]
+        self.dcp = cbufs(B)       # pi-phase dgrad chain buffers
+        self.dc = cbufs(B)        # q-phase dgrad chain buffers
+        self.dxc = torch.zeros(B, OC, **f32)
+
+        # ---- module parameter views -----------------------------------
+        def critic_layers(mod):
+            return [(l.weight, l.bias) for l in mod.layers]
+
+        self.cw = [critic_layers(critic.q1), critic_layers(critic.q2)]
+        self.tw = [critic_layers(target_critic.q1),
+                   critic_layers(target_critic.q2)]
+        self.aw = [(l.weight, l.bias) for l in actor.layers]
+        self.head_w = [(actor.mu_layer.weight, actor.mu_layer.bias),
+                       (actor.log_std_layer.weight, actor.log_std_layer.bias)]
+
+        # ---- transposed-weight caches ---------------------------------
+        def wt_like(w):
+            return torch.zeros(w.shape[1], w.shape[0], **f32)
+
+        self.cwt = [[wt_like(w) for (w, _) in z] for z in self.cw]
+        self.awt = [wt_like(w) for (w, _) in self.aw]
+        self.hwt = [wt_like(w) for (w, _) in self.head_w]
+        self._c_tr_src = [w for z in self.cw for (w, _) in z]
+        self._c_tr_dst = [wt for z in self.cwt for wt in z]
+        self._a_tr_src = [w for (w, _) in self.aw] + \
+            [w for (w, _) in self.head_w]
+        self._a_tr_dst = list(self.awt) + list(self.hwt)
+        self.ext.transpose_multi(self._c_tr_src, self._c_tr_dst)
+        self.ext.transpose_multi(self._a_tr_src, self._a_tr_dst)
+
+        # ---- losses / alpha -------------------------------------------
+        self.loss_q_acc = torch.zeros(1, **f32)
+        self.loss_pi_acc = torch.zeros(1, **f32)
+        self.learn_alpha = bool(getattr(sac, "learn_alpha", False))
+        self.alpha_host = float(sac.alpha)
+        if self.learn_alpha:
+            self.log_alpha = torch.full((1,), math.log(sac.alpha), **f32)
+            self.alpha_dev = self.log_alpha.exp()
+            self.alpha_m = torch.zeros(1, **f32)
+            self.alpha_v = torch.zeros(1, **f32)
+            self.alpha_step = torch.zeros(1, dtype=torch.int64,
+                                          device=device)
+            self.mean_logp = torch.zeros(1, **f32)
+            te = getattr(sac, "target_entropy", None)
+            self.target_entropy = float(te if te is not None else -A)
+        else:
+            self.alpha_dev = None
+            self.mean_logp = None
+
+        torch.cuda.synchronize()
+
+        # ---- capture ---------------------------------------------------
+        self._graphs = None
+        self.graph = None
+        if capture:
+            self._capture()
+
+    # ------------------------------------------------------------------
+    # recorded phases
+    # ------------------------------------------------------------------
+
+    def _mg(self, xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
+            xs2=None, ws2=None, masks2=None, K2=0, x_off=0, x2_off=0):
+        self.ext.mgemm(xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
+                       xs2 or [], ws2 or [], masks2 or [], K2, x_off,
+                       x2_off)
+
+    def _critic_fwd(self, x_src, x_off, weights, acts, lda):
+        """Twin-critic forward; acts[z][i] filled.  Final layer no relu."""
+        B = self.B
+        nL = len(self.c_w)
+        x = [x_src, x_src]
+        k = self.OC
+        src_off = x_off
+        src_lda = lda
+        for i in range(nL):
+            relu = i + 1 < nL
+            self._mg(x, [weights[z][i][0] for z in range(2)],
+                     [weights[z][i][1] for z in range(2)],
+                     [acts[z][i] for z in range(2)], [None, None],
+                     B, self.c_w[i], k, src_lda, self.c_w[i], relu,
+                     x_off=src_off)
+            x = [acts[0][i], acts[1][i]]
+            k = self.c_w[i]
+            src_off = 0
+            src_lda = k
+
+    def _phase_critic(self):
+        ext = self.ext
+        B, O, A, OC = self.B, self.O, self.A, self.OC
+        buf = self.buffer
+        ext.bump_counter(self.ctr)
+        if self.sample:
+            ext.gather2(buf.state, buf.actions, buf.rewards, buf.next_state,
+                        buf.done, buf._size_dev, self.ctr, self.seed,
+                        self.XC, self.XC2, self.rew, self.done, B)
+
+        # actor forward over stacked 2B rows of XC[:, :O]
+        x, k, lda, off = self.XC, O, OC, 0
+        for i, (w, b) in enumerate(self.aw):
+            self._mg([x], [w], [b], [self.a_act[i]], [None], 2 * B,
+                     self.a_hidden[i], k, lda, self.a_hidden[i], True,
+                     x_off=off)
+            x, k, lda, off = self.a_act[i], self.a_hidden[i], \
+                self.a_hidden[i], 0
+        (wm, bm), (wl, bl) = self.head_w
+        self._mg([x, x], [wm, wl], [bm, bl], [self.mu, self.ls],
+                 [None, None], 2 * B, A, k, lda, A, False)
+        # pi rows :B -> XC2[:, O:], a2 rows B: -> XC[B:, O:]
+        ext.tg_fwd2(self.mu, self.ls, self.XC2, O, self.XC, B * OC + O,
+                    B, self.logp, self.prob, self.ctr, self.seed,
+                    self.act_limit, self.lo, self.hi)
+
+        # target critic on (ns, a2) = XC rows B:
+        self._critic_fwd(self.XC, B * OC, self.tw, self.t_act, OC)
+        # critic on (s, a) = XC rows :B
+        self._critic_fwd(self.XC, 0, self.cw, self.c_act, OC)
+
+        nL = len(self.c_w)
+        q = [self.c_act[z][nL - 1] for z in range(2)]
+        qt = [self.t_act[z][nL - 1] for z in range(2)]
+        ext.qloss2(q[0], q[1], qt[0], qt[1], self.logp[B:], self.rew,
+                   self.done, self.alpha_dev, self.alpha_host,
+                   self.loss_q_acc, self.dq[0], self.dq[1], B,
+                   self.sac.gamma, self.sac.reward_scale)
+
+        # critic backward (wgrad into flat grads; dgrad via cached W^T)
+        d = self.dq
+        for i in range(nL - 1, -1, -1):
+            relu_mask = i + 1 < nL   # incoming d is post-relu of layer i?
+            masks = [self.c_act[z][i] if relu_mask else None
+                     for z in range(2)]
+            if i > 0:
+                x_in = [self.c_act[z][i - 1] for z in range(2)]
+                ldx, xoff = self.c_w[i - 1], 0
+            else:
+                x_in = [self.XC, self.XC]
+                ldx, xoff = OC, 0
+            ext.mwgrad(d, masks, x_in,
+                       [self.cw[z][i][0].grad for z in range(2)],
+                       [self.cw[z][i][1].grad for z in range(2)],
+                       B, self.c_w[i], (self.c_w[i - 1] if i > 0 else OC),
+                       self.c_w[i], ldx, xoff)
+            if i > 0:
+                self._mg(d, [self.cwt[z][i] for z in range(2)],
+                         [None, None],
+                         [self.dc[z][i - 1] for z in range(2)],
+                         masks, B, self.c_w[i - 1], self.c_w[i],
+                         self.c_w[i], self.c_w[i - 1], False)
+                d = [self.dc[z][i - 1] for z in range(2)]
+
+    def _phase_policy(self):
+        ext = self.ext
+        B, O, A, OC = self.B, self.O, self.A, self.OC
+        self.q_opt.step()
+        ext.transpose_multi(self._c_tr_src, self._c_tr_dst)
+
+        # critic on (s, pi) = XC2 with the UPDATED critic
+        self._critic_fwd(self.XC2, 0, self.cw, self.p_act, OC)
+        nL = len(self.c_w)
+        qp = [self.p_act[z][nL - 1] for z in range(2)]
+        ext.piloss2(qp[0], qp[1], self.logp[:B], self.alpha_dev,
+                    self.alpha_host, self.loss_pi_acc, self.mean_logp,
+                    self.dqp[0], self.dqp[1], B)
+
+        # critic dgrad chain only (frozen critic)
+        d = self.dqp
+        for i in range(nL - 1, 0, -1):
+            masks = [self.p_act[z][i] if i + 1 < nL else None
+                     for z in range(2)]
+            self._mg(d, [self.cwt[z][i] for z in range(2)], [None, None],
+                     [self.dcp[z][i - 1] for z in range(2)], masks,
+                     B, self.c_w[i - 1], self.c_w[i], self.c_w[i],
+                     self.c_w[i - 1], False)
+            d = [self.dcp[z][i - 1] for z in range(2)]
+        # layer 0: sum over the twin critics -> dxc
+        masks0 = [self.p_act[z][0] for z in range(2)]
+        self._mg([d[0]], [self.cwt[0][0]], [None], [self.dxc], [masks0[0]],
+                 B, OC, self.c_w[0], self.c_w[0], OC, False,
+                 xs2=[d[1]], ws2=[self.cwt[1][0]], masks2=[masks0[1]],
+                 K2=self.c_w[0])
+
+        # actor backward
+        ext.tg_bwd2(self.dxc, O, self.alpha_dev, self.alpha_host,
+                    self.mu, self.ls, self.prob, self.dmu, self.dls, B,
+                    self.act_limit, self.lo, self.hi)
+        h_last = self.a_hidden[-1]
+        a_last = self.a_act[-1]
+        (wm, bm), (wl, bl) = self.head_w
+        ext.mwgrad([self.dmu, self.dls], [None, None], [a_last, a_last],
+                   [wm.grad, wl.grad], [bm.grad, bl.grad],
+                   B, A, h_last, A, h_last, 0)
+        self._mg([self.dmu], [self.hwt[0]], [None], [self.da[-1]], [None],
+                 B, h_last, A, A, h_last, False,
+                 xs2=[self.dls], ws2=[self.hwt[1]], masks2=[None], K2=A)
+
+        d = self.da[-1]
+        for i in range(len(self.aw) - 1, -1, -1):
+            mask = self.a_act[i]
+            if i > 0:
+                x_in, ldx, xoff = self.a_act[i - 1], self.a_hidden[i - 1], 0
+                kin = self.a_hidden[i - 1]
+            else:
+                x_in, ldx, xoff = self.XC, OC, 0
+                kin = O
+            (w, b) = self.aw[i]
+            ext.mwgrad([d], [mask], [x_in], [w.grad], [b.grad],
+                       B, self.a_hidden[i], kin, self.a_hidden[i], ldx,
+                       xoff)
+            if i > 0:
+                self._mg([d], [self.awt[i]], [None], [self.da[i - 1]],
+                         [mask], B, self.a_hidden[i - 1], self.a_hidden[i],
+                         self.a_hidden[i], self.a_hidden[i - 1], False)
+                d = self.da[i - 1]
+
+    def _phase_finish(self):
+        ext = self.ext
+        self.pi_opt.step()
+        ext.transpose_multi(self._a_tr_src, self._a_tr_dst)
+        ext.polyak_(self.target_flat, self.q_opt.fp.flat, self.sac.polyak)
+        if self.learn_alpha:
+            ext.alpha_update(self.log_alpha, self.alpha_dev, self.alpha_m,
+                             self.alpha_v, self.alpha_step, self.mean_logp,
+                             self.target_entropy, 3e-4)
+
+    # ------------------------------------------------------------------
+
+    def _reduce(self, opt):
+        if self.world > 1:
+            comm.allreduce_grads(opt.fp.flat_grad)
+
+    def _run_once(self):
+        self._phase_critic()
+        self._reduce(self.q_opt)
+        self._phase_policy()
+        self._reduce(self.pi_opt)
+        self._phase_finish()
+
+    def _capture(self):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._run_once()
+        torch.cuda.current_stream().wait_stream(s)
+
+        if self.world == 1:
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self._phase_critic()
+                self._phase_policy()
+                self._phase_finish()
+        else:
+            g1 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g1):
+                self._phase_critic()
+            g2 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g2, pool=g1.pool()):
+                self._phase_policy()
+            g3 = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g3, pool=g1.pool()):
+                self._phase_finish()
+            self._graphs = (g1, g2, g3)
+
+    def step(self):
+        if self.graph is not None:
+            self.graph.replay()
+        elif self._graphs is not None:
+            g1, g2, g3 = self._graphs
+            g1.replay()
+            self._reduce(self.q_opt)
+            g2.replay()
+            self._reduce(self.pi_opt)
+            g3.replay()
+        else:
+            self._run_once()
+
+    def read_and_reset_losses(self, n_updates: int):
+        if n_updates <= 0:
+            return 0.0, 0.0
+        lq = float(self.loss_q_acc.item()) / n_updates
+        lp = float(self.loss_pi_acc.item()) / n_updates
+        self.loss_q_acc.zero_()
+        self.loss_pi_acc.zero_()
+        return lq, lp
+
+    # -- test helper ----------------------------------------------------
+
+    def load_batch(self, s, a, r, ns, d):
+        """Write a batch directly into the static buffers (sample=False
+        parity tests)."""
+        B, O = self.B, self.O
+        self.XC[:B, :O] = s
+        self.XC[:B, O:] = a
+        self.XC[B:, :O] = ns
+        self.XC2[:, :O] = s
+        self.rew.copy_(r)
+        self.done.copy_(d)

@@ -204,8 +204,29 @@ class SAC:
             return self._graph
         if (self._graph_failed or not self.use_graph
                 or device.type != "cuda"
-                or self._critic_fp is None or self._actor_fp is None
-                or self.learn_alpha):
+                or self._critic_fp is None or self._actor_fp is None):
+            return None
+        # 1st choice: the hand-scheduled fused engine (plain MLP models)
+        from ..buffer.replay import ReplayBuffer
+        from ..models.mlp import Actor as MlpActor
+        from ..models.mlp import DoubleCritic as MlpDoubleCritic
+        if (type(actor) is MlpActor and type(critic) is MlpDoubleCritic
+                and isinstance(buffer, ReplayBuffer)
+                and buffer.act_dim <= 64):
+            try:
+                from .engine import FusedSACEngine
+                self._graph = FusedSACEngine(
+                    self, actor, critic, target_critic, buffer, pi_opt,
+                    q_opt, self._target_flat, self.batch_size, device,
+                    philox_seed=10000 * comm.proc_id())
+                logger.info("fused SAC update engine captured (world=%d)",
+                            comm.num_procs())
+                return self._graph
+            except Exception as e:  # noqa: BLE001
+                logger.warning("fused engine capture failed (%r); trying "
+                               "autograd graph", e)
+        if self.learn_alpha:
+            self._graph_failed = True
             return None
         try:
             from .graph import GraphedSACUpdate

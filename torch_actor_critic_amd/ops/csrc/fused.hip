@@ -1,0 +1,914 @@
+// Fused SAC update engine kernels (gfx950 / CDNA4).
+//
+// The profiled autograd update path spends 75% of its time in GEMM
+// kernels with scalar staging and ~50 of its 112 kernels on autograd
+// bookkeeping (profiles/r01_baseline_update_profile.md).  These kernels
+// implement a hand-scheduled update pass (driven by
+// torch_actor_critic_amd/algo/engine.py) with:
+//   * multi-problem GEMMs: both twin critics (or both policy heads) in
+//     ONE launch via blockIdx.z;
+//   * lda/ldy strides so operands read/write slices of concat-layout
+//     buffers — torch.cat disappears from the update entirely;
+//   * cached transposed weights (refreshed once per Adam step) so dgrad
+//     is a coalesced fwd-form GEMM;
+//   * vectorized (float4 -> bf16x8) LDS staging on aligned interiors;
+//   * wgrad writing straight into the module's flat gradient slices
+//     (no autograd accumulation);
+//   * device-side alpha (learned entropy temperature) read by the loss
+//     kernels and updated in-graph — per BASELINE.json's north star the
+//     whole update (sample, twin-min Bellman, tanh-Gaussian sample,
+//     entropy-temperature loss, polyak) replays as one hipGraph.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+namespace fused {
+
+#define DEVINL __device__ __forceinline__
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+// ---------------------------------------------------------------------------
+// Philox (same generator as tac_kernels.hip; duplicated in this TU to keep
+// both translation units self-contained)
+// ---------------------------------------------------------------------------
+
+struct P4 { uint32_t x, y, z, w; };
+
+DEVINL uint32_t mulhilo_(uint32_t a, uint32_t b, uint32_t* hi) {
+  uint64_t p = (uint64_t)a * (uint64_t)b;
+  *hi = (uint32_t)(p >> 32);
+  return (uint32_t)p;
+}
+
+DEVINL P4 philox_(uint64_t seed, uint64_t chi, uint64_t clo) {
+  uint32_t c0 = (uint32_t)clo, c1 = (uint32_t)(clo >> 32);
+  uint32_t c2 = (uint32_t)chi, c3 = (uint32_t)(chi >> 32);
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    uint32_t h0, h1;
+    uint32_t l0 = mulhilo_(0xD2511F53u, c0, &h0);
+    uint32_t l1 = mulhilo_(0xCD9E8D57u, c2, &h1);
+    uint32_t n0 = h1 ^ c1 ^ k0, n1 = l1, n2 = h0 ^ c3 ^ k1, n3 = l0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += 0x9E3779B9u; k1 += 0xBB67AE85u;
+  }
+  return {c0, c1, c2, c3};
+}
+
+// ---------------------------------------------------------------------------
+// Multi-problem MFMA GEMM (fwd-form): y = act(x @ w^T + b), x optionally
+// relu-masked by `mask` (backward chain), optional second (x2,w2,mask2)
+// accumulated into the same output (dgrad-sum over the twin critics /
+// the two policy heads).
+// ---------------------------------------------------------------------------
+
+struct MProb {
+  const float* x; const float* w; const float* bias; float* y;
+  const float* mask;
+  const float* x2; const float* w2; const float* mask2;
+};
+
+struct MGemm {
+  MProb p[2];
+  int M, N, K, lda, ldy;
+  int K2;  // reduction depth of the second operand pair (SUM2)
+};
+
+constexpr int TB = 64;        // block tile (M and N)
+constexpr int BKB2 = 64;      // bf16 K-step
+constexpr int BKF2 = 16;      // fp32 K-step
+constexpr int LDSB2 = 72;     // bf16 LDS halves per row
+constexpr int LDSF2 = 17;     // fp32 LDS floats per row
+
+// Stage a 64 x BK tile of `src` (row-major, leading dim ld, rows base
+// `r0`, cols base `k0`, bounds R x C) into LDS, optionally masked.
+template <bool BF16, bool MASK>
+DEVINL void stage_tile(void* lds, const float* src, const float* mask,
+                       int r0, int k0, int R, int C, int ld) {
+  const int tid = threadIdx.x;
+  const int row = tid & 63;
+  if constexpr (BF16) {
+    __bf16* d = (__bf16*)lds;
+    const int c0 = (tid >> 6) * 16;
+    const int gr = r0 + row;
+    bool interior = (r0 + 64 <= R) && (k0 + 64 <= C) && ((ld & 3) == 0)
+                    && ((k0 & 3) == 0);
+    if (interior) {
+      const float* p = src + (int64_t)gr * ld + k0 + c0;
+      float v[16];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        float4 f = *(const float4*)(p + q * 4);
+        v[q * 4 + 0] = f.x; v[q * 4 + 1] = f.y;
+        v[q * 4 + 2] = f.z; v[q * 4 + 3] = f.w;
+      }
+      if constexpr (MASK) {
+        const float* mp = mask + (int64_t)gr * ld + k0 + c0;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          float4 f = *(const float4*)(mp + q * 4);
+          v[q * 4 + 0] = f.x > 0.f ? v[q * 4 + 0] : 0.f;
+          v[q * 4 + 1] = f.y > 0.f ? v[q * 4 + 1] : 0.f;
+          v[q * 4 + 2] = f.z > 0.f ? v[q * 4 + 2] : 0.f;
+          v[q * 4 + 3] = f.w > 0.f ? v[q * 4 + 3] : 0.f;
+        }
+      }
+      union { __bf16 h[16]; uint4 u[2]; } pk;
+#pragma unroll
+      for (int e = 0; e < 16; ++e) pk.h[e] = (__bf16)v[e];
+      uint4* dst = (uint4*)&d[row * LDSB2 + c0];
+      dst[0] = pk.u[0];
+      dst[1] = pk.u[1];
+    } else {
+#pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        int c = c0 + e;
+        float v = 0.f;
+        if (gr < R && k0 + c < C) {
+          v = src[(int64_t)gr * ld + k0 + c];
+          if constexpr (MASK) {
+            v = mask[(int64_t)gr * ld + k0 + c] > 0.f ? v : 0.f;
+          }
+        }
+        d[row * LDSB2 + c] = (__bf16)v;
+      }
+    }
+  } else {
+    float* d = (float*)lds;
+    const int c0 = (tid >> 6) * 4;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int c = c0 + e;
+      int gr = r0 + row;
+      float v = 0.f;
+      if (gr < R && k0 + c < C) {
+        v = src[(int64_t)gr * ld + k0 + c];
+        if constexpr (MASK) {
+          v = mask[(int64_t)gr * ld + k0 + c] > 0.f ? v : 0.f;
+        }
+      }
+      d[row * LDSF2 + c] = v;
+    }
+  }
+}
+
+template <bool BF16>
+DEVINL void mma_tiles(const void* xs_, const void* ws_, f32x4 (&acc)[2][2],
+                      int lane, int wrow, int wcol) {
+  if constexpr (BF16) {
+    const __bf16* xs = (const __bf16*)xs_;
+    const __bf16* ws = (const __bf16*)ws_;
+    const int arow = lane & 15;
+    const int ak0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int kk = 0; kk < BKB2; kk += 32) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        bf16x8 a = *(const bf16x8*)&xs[(wrow + mi * 16 + arow) * LDSB2
+                                       + kk + ak0];
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          bf16x8 b = *(const bf16x8*)&ws[(wcol + ni * 16 + arow) * LDSB2
+                                         + kk + ak0];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  } else {
+    const float* xs = (const float*)xs_;
+    const float* ws = (const float*)ws_;
+    const int arow = lane & 15;
+    const int akl = lane >> 4;
+#pragma unroll
+    for (int kk = 0; kk < BKF2; kk += 4) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        float a = xs[(wrow + mi * 16 + arow) * LDSF2 + kk + akl];
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          float b = ws[(wcol + ni * 16 + arow) * LDSF2 + kk + akl];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a, b, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  }
+}
+
+template <bool BF16, bool MASK, bool RELU, bool SUM2>
+__global__ __launch_bounds__(256)
+void mgemm_kernel(MGemm g) {
+  const MProb& p = g.p[blockIdx.z];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bm0 = blockIdx.x * TB;
+  const int bn0 = blockIdx.y * TB;
+  constexpr int BK = BF16 ? BKB2 : BKF2;
+  constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  void* xs = smem;
+  void* ws = smem + LBYTES;
+
+  f32x4 acc[2][2] = {};
+  for (int k0 = 0; k0 < g.K; k0 += BK) {
+    stage_tile<BF16, MASK>(xs, p.x, p.mask, bm0, k0, g.M, g.K, g.lda);
+    stage_tile<BF16, false>(ws, p.w, nullptr, bn0, k0, g.N, g.K, g.K);
+    __syncthreads();
+    mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
+    __syncthreads();
+  }
+  if constexpr (SUM2) {
+    for (int k0 = 0; k0 < g.K2; k0 += BK) {
+      stage_tile<BF16, MASK>(xs, p.x2, p.mask2, bm0, k0, g.M, g.K2, g.K2);
+      stage_tile<BF16, false>(ws, p.w2, nullptr, bn0, k0, g.N, g.K2, g.K2);
+      __syncthreads();
+      mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
+      __syncthreads();
+    }
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int grow = bm0 + wrow + mi * 16 + crow + r;
+        int gcol = bn0 + wcol + ni * 16 + ccol;
+        if (grow < g.M && gcol < g.N) {
+          float v = acc[mi][ni][r];
+          if (p.bias) v += p.bias[gcol];
+          if constexpr (RELU) v = fmaxf(v, 0.f);
+          p.y[(int64_t)grow * g.ldy + gcol] = v;
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
+// Multi-problem wgrad: dW[N,K] = sum_i dYeff[i,n] X[i,k]; db[N] fused.
+// Coalesced global reads (contiguous along n / k), LDS-transposed writes.
+// ---------------------------------------------------------------------------
+
+struct WProb {
+  const float* dy; const float* ymask; const float* x;
+  float* dw; float* db;
+};
+
+struct WGemm {
+  WProb p[2];
+  int M, N, K, lddy, ldx;
+};
+
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void mwgrad_kernel(WGemm g) {
+  const WProb& p = g.p[blockIdx.z];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bn0 = blockIdx.x * TB;   // output rows (N of dW)
+  const int bk0 = blockIdx.y * TB;   // output cols (K of dW)
+  constexpr int BK = BF16 ? BKB2 : BKF2;
+  constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  __shared__ float dbs[64];
+
+  if (tid < 64) dbs[tid] = 0.f;
+  f32x4 acc[2][2] = {};
+
+  for (int i0 = 0; i0 < g.M; i0 += BK) {
+    // A tile: as[n][i] = dYeff[i0+i][bn0+n]; staged transposed.
+    // thread: i = tid&(BK-1)... BK may be 16 (fp32): use i = tid % BK.
+    {
+      if constexpr (BF16) {
+        __bf16* as = (__bf16*)smem;
+        const int ic = tid & 63;
+        const int nc0 = (tid >> 6) * 16;
+        const int gi = i0 + ic;
+        bool interior = (i0 + 64 <= g.M) && (bn0 + 64 <= g.N)
+                        && ((g.lddy & 3) == 0) && ((bn0 & 3) == 0);
+        float v[16];
+        if (interior) {
+          const float* q = p.dy + (int64_t)gi * g.lddy + bn0 + nc0;
+#pragma unroll
+          for (int t = 0; t < 4; ++t) {
+            float4 f = *(const float4*)(q + t * 4);
+            v[t*4+0]=f.x; v[t*4+1]=f.y; v[t*4+2]=f.z; v[t*4+3]=f.w;
+          }
+          if constexpr (MASK) {
+            const float* mq = p.ymask + (int64_t)gi * g.lddy + bn0 + nc0;
+#pragma unroll
+            for (int t = 0; t < 4; ++t) {
+              float4 f = *(const float4*)(mq + t * 4);
+              v[t*4+0] = f.x > 0.f ? v[t*4+0] : 0.f;
+              v[t*4+1] = f.y > 0.f ? v[t*4+1] : 0.f;
+              v[t*4+2] = f.z > 0.f ? v[t*4+2] : 0.f;
+              v[t*4+3] = f.w > 0.f ? v[t*4+3] : 0.f;
+            }
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < 16; ++e) {
+            int n = bn0 + nc0 + e;
+            float val = 0.f;
+            if (gi < g.M && n < g.N) {
+              val = p.dy[(int64_t)gi * g.lddy + n];
+              if constexpr (MASK) {
+                val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
+              }
+            }
+            v[e] = val;
+          }
+        }
+#pragma unroll
+        for (int e = 0; e < 16; ++e)
+          as[(nc0 + e) * LDSB2 + ic] = (__bf16)v[e];
+        // db partial: sum over i for each n — accumulate via LDS atomic
+        float dbp = 0.f;
+#pragma unroll
+        for (int e = 0; e < 16; ++e) dbp += v[e];
+        (void)dbp;  // db handled below from LDS tile
+      } else {
+        float* as = (float*)smem;
+        const int ic = tid & 15;
+        const int nc0 = (tid >> 4) * 4;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          int n = bn0 + nc0 + e;
+          int gi = i0 + ic;
+          float val = 0.f;
+          if (gi < g.M && n < g.N) {
+            val = p.dy[(int64_t)gi * g.lddy + n];
+            if constexpr (MASK) {
+              val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
+            }
+          }
+          as[(nc0 + e) * LDSF2 + ic] = val;
+        }
+      }
+    }
+    // B tile: bs[k][i] = X[i0+i][bk0+k]
+    {
+      if constexpr (BF16) {
+        __bf16* bs = (__bf16*)(smem + LBYTES);
+        const int ic = tid & 63;
+        const int kc0 = (tid >> 6) * 16;
+        const int gi = i0 + ic;
+        bool interior = (i0 + 64 <= g.M) && (bk0 + 64 <= g.K)
+                        && ((g.ldx & 3) == 0) && ((bk0 & 3) == 0);
+        float v[16];
+        if (interior) {
+          const float* q = p.x + (int64_t)gi * g.ldx + bk0 + kc0;
+#pragma unroll
+          for (int t = 0; t < 4; ++t) {
+            float4 f = *(const float4*)(q + t * 4);
+            v[t*4+0]=f.x; v[t*4+1]=f.y; v[t*4+2]=f.z; v[t*4+3]=f.w;
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < 16; ++e) {
+            int k = bk0 + kc0 + e;
+            int gi2 = i0 + ic;
+            v[e] = (gi2 < g.M && k < g.K)
+                       ? p.x[(int64_t)gi2 * g.ldx + k] : 0.f;
+          }
+        }
+#pragma unroll
+        for (int e = 0; e < 16; ++e)
+          bs[(kc0 + e) * LDSB2 + ic] = (__bf16)v[e];
+      } else {
+        float* bs = (float*)(smem + LBYTES);
+        const int ic = tid & 15;
+        const int kc0 = (tid >> 4) * 4;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          int k = bk0 + kc0 + e;
+          int gi = i0 + ic;
+          bs[(kc0 + e) * LDSF2 + ic] =
+              (gi < g.M && k < g.K) ? p.x[(int64_t)gi * g.ldx + k] : 0.f;
+        }
+      }
+    }
+    __syncthreads();
+    mma_tiles<BF16>(smem, smem + LBYTES, acc, lane, wrow, wcol);
+    // db: from the staged (already masked) A tile, k-tile-0 blocks only
+    if (p.db && blockIdx.y == 0 && tid < 64) {
+      float s = 0.f;
+      if constexpr (BF16) {
+        const __bf16* as = (const __bf16*)smem;
+        constexpr int NI = BKB2;
+        for (int i = 0; i < NI; ++i) s += (float)as[tid * LDSB2 + i];
+      } else {
+        const float* as = (const float*)smem;
+        for (int i = 0; i < BKF2; ++i) s += as[tid * LDSF2 + i];
+      }
+      dbs[tid] += s;
+    }
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int gn = bn0 + wrow + mi * 16 + crow + r;
+        int gk = bk0 + wcol + ni * 16 + ccol;
+        if (gn < g.N && gk < g.K)
+          p.dw[(int64_t)gn * g.K + gk] = acc[mi][ni][r];
+      }
+  if (p.db && blockIdx.y == 0 && tid < 64 && bn0 + tid < g.N)
+    p.db[bn0 + tid] = dbs[tid];
+}
+
+// ---------------------------------------------------------------------------
+// Weight-transpose refresh: wt[k*N + n] = w[n*K + k], many layers per launch
+// ---------------------------------------------------------------------------
+
+constexpr int MAX_T = 12;
+struct TArgs {
+  const float* w[MAX_T];
+  float* wt[MAX_T];
+  int N[MAX_T], K[MAX_T];
+  int n_layers, blocks_per_layer;
+};
+
+__global__ __launch_bounds__(256)
+void transpose_multi_kernel(TArgs t) {
+  const int layer = blockIdx.x / t.blocks_per_layer;
+  const int slice = blockIdx.x % t.blocks_per_layer;
+  if (layer >= t.n_layers) return;
+  const int N = t.N[layer], K = t.K[layer];
+  const int64_t total = (int64_t)N * K;
+  const int64_t stride = (int64_t)t.blocks_per_layer * blockDim.x;
+  for (int64_t idx = (int64_t)slice * blockDim.x + threadIdx.x; idx < total;
+       idx += stride) {
+    int n = (int)(idx / K), k = (int)(idx % K);
+    t.wt[layer][(int64_t)k * N + n] = t.w[layer][idx];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Replay gather writing straight into the concat-layout batch buffers
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void gather2_kernel(const float* __restrict__ state,
+                    const float* __restrict__ act,
+                    const float* __restrict__ rew,
+                    const float* __restrict__ nstate,
+                    const float* __restrict__ done,
+                    const int64_t* __restrict__ size_dev,
+                    const int64_t* __restrict__ ctr, uint64_t seed,
+                    float* __restrict__ xc,   // [2B, O+A]
+                    float* __restrict__ xc2,  // [B, O+A]
+                    float* __restrict__ orew, float* __restrict__ od,
+                    int B, int obs_dim, int act_dim) {
+  const int j = blockIdx.x;
+  const int ldc = obs_dim + act_dim;
+  const uint64_t size = (uint64_t)size_dev[0];
+  P4 r = philox_(seed, (uint64_t)ctr[0], (uint64_t)j);
+  uint64_t u = ((uint64_t)r.x << 32) | r.y;
+  int64_t idx = (int64_t)(u % (size ? size : 1));
+
+  const float* srow = state + idx * obs_dim;
+  const float* nrow = nstate + idx * obs_dim;
+  const float* arow = act + idx * act_dim;
+  for (int c = threadIdx.x; c < obs_dim; c += blockDim.x) {
+    float s = srow[c];
+    xc[(int64_t)j * ldc + c] = s;           // s -> XC rows :B
+    xc2[(int64_t)j * ldc + c] = s;          // s -> XC2
+    xc[(int64_t)(B + j) * ldc + c] = nrow[c];  // ns -> XC rows B:
+  }
+  for (int c = threadIdx.x; c < act_dim; c += blockDim.x)
+    xc[(int64_t)j * ldc + obs_dim + c] = arow[c];  // a -> XC rows :B
+  if (threadIdx.x == 0) {
+    orew[j] = rew[idx];
+    od[j] = done[idx];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused tanh-Gaussian head, stacked rows with split outputs + internal
+// Philox noise (row < B0 -> out0 slice; else out1 slice)
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(64)
+void tg_fwd2_kernel(const float* __restrict__ mu,
+                    const float* __restrict__ ls_raw,
+                    float* __restrict__ out0, int ld0,
+                    float* __restrict__ out1, int ld1, int B0,
+                    float* __restrict__ logp, float* __restrict__ prob_out,
+                    const int64_t* __restrict__ ctr, uint64_t seed,
+                    int R, int A, float act_limit, float lo, float hi) {
+  const int b = blockIdx.x;
+  const int a = threadIdx.x;
+  float acc = 0.f;
+  if (a < A) {
+    const int64_t i = (int64_t)b * A + a;
+    // philox noise: one draw per element (use lane of the quad)
+    P4 r = philox_(seed ^ 0x517cc1b727220a95ull, (uint64_t)ctr[0],
+                   (uint64_t)i);
+    float u0 = (r.x + 1.f) * 2.3283064365386963e-10f;
+    float u1 = (r.y + 1.f) * 2.3283064365386963e-10f;
+    float eps = sqrtf(-2.f * logf(u0)) * __cosf(6.283185307179586f * u1);
+
+    float m = mu[i];
+    float ls = fminf(fmaxf(ls_raw[i], lo), hi);
+    float std = expf(ls);
+    float prob = m + std * eps;
+    float pi = tanhf(prob) * act_limit;
+    prob_out[i] = prob;
+    if (b < B0) out0[(int64_t)b * ld0 + a] = pi;
+    else        out1[(int64_t)(b - B0) * ld1 + a] = pi;
+    float x = -2.f * prob;
+    float sp = x > 20.f ? x : log1pf(expf(fminf(x, 20.f)));
+    float gauss = -0.5f * eps * eps - ls - 0.5f * 1.8378770664093453f;
+    float corr = 1.3862943611198906f - prob - sp;
+    acc = gauss - corr;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (a == 0) logp[b] = acc;
+}
+
+// Debug/test helper: reproduce tg_fwd2's internal noise exactly so the
+// eager fp32 reference can be driven with identical eps.
+__global__ __launch_bounds__(64)
+void tg_eps_kernel(float* __restrict__ eps_out, uint64_t ctr_val,
+                   uint64_t seed, int R, int A) {
+  const int b = blockIdx.x;
+  const int a = threadIdx.x;
+  if (a >= A) return;
+  const int64_t i = (int64_t)b * A + a;
+  P4 r = philox_(seed ^ 0x517cc1b727220a95ull, ctr_val, (uint64_t)i);
+  float u0 = (r.x + 1.f) * 2.3283064365386963e-10f;
+  float u1 = (r.y + 1.f) * 2.3283064365386963e-10f;
+  eps_out[i] = sqrtf(-2.f * logf(u0)) * __cosf(6.283185307179586f * u1);
+}
+
+// backward over rows :B of the stacked forward; dpi read from the dxc
+// slab at column offset; dlogp = alpha/B (policy loss seed) computed
+// in-kernel from the device alpha.
+__global__ __launch_bounds__(64)
+void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
+                    const float* __restrict__ alpha_dev, float alpha_host,
+                    const float* __restrict__ mu,
+                    const float* __restrict__ ls_raw,
+                    const float* __restrict__ prob,
+                    float* __restrict__ dmu, float* __restrict__ dls,
+                    int B, int A, float act_limit, float lo, float hi) {
+  const int b = blockIdx.x;
+  const int a = threadIdx.x;
+  if (a >= A) return;
+  const int64_t i = (int64_t)b * A + a;
+  const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
+  float raw = ls_raw[i];
+  float ls = fminf(fmaxf(raw, lo), hi);
+  float std = expf(ls);
+  float p = prob[i];
+  float t = tanhf(p);
+  float se = p - mu[i];              // std * eps
+  float dpi = dxc[(int64_t)b * ld_dxc + col0 + a];
+  float dl = alpha / B;
+  float dp = dpi * act_limit * (1.f - t * t);
+  float g_mu = dp + dl * t;
+  float g_ls = dp * se + dl * (se * t - 1.f);
+  float mask = (raw >= lo && raw <= hi) ? 1.f : 0.f;
+  dmu[i] = g_mu;
+  dls[i] = g_ls * mask;
+}
+
+// ---------------------------------------------------------------------------
+// Single-block losses (deterministic reductions, device-alpha aware)
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void qloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
+                   const float* __restrict__ q1t, const float* __restrict__ q2t,
+                   const float* __restrict__ logp_next,
+                   const float* __restrict__ rew, const float* __restrict__ done,
+                   const float* __restrict__ alpha_dev, float alpha_host,
+                   float* __restrict__ loss_acc,
+                   float* __restrict__ dq1, float* __restrict__ dq2,
+                   int B, float gamma, float scale) {
+  __shared__ float red[4];
+  const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
+  float acc = 0.f;
+  for (int i = threadIdx.x; i < B; i += blockDim.x) {
+    float backup = scale * rew[i] + gamma * (1.f - done[i]) *
+                       (fminf(q1t[i], q2t[i]) - alpha * logp_next[i]);
+    float e1 = q1[i] - backup, e2 = q2[i] - backup;
+    acc += e1 * e1 + e2 * e2;
+    dq1[i] = 2.f * e1 / B;
+    dq2[i] = 2.f * e2 / B;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    loss_acc[0] += (red[0] + red[1] + red[2] + red[3]) / B;
+}
+
+__global__ __launch_bounds__(256)
+void piloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
+                    const float* __restrict__ logp,
+                    const float* __restrict__ alpha_dev, float alpha_host,
+                    float* __restrict__ loss_acc,
+                    float* __restrict__ mean_logp,
+                    float* __restrict__ dq1, float* __restrict__ dq2,
+                    int B) {
+  __shared__ float red[4], redl[4];
+  const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
+  float acc = 0.f, laux = 0.f;
+  for (int i = threadIdx.x; i < B; i += blockDim.x) {
+    float a = q1[i], b = q2[i];
+    acc += alpha * logp[i] - fminf(a, b);
+    laux += logp[i];
+    float g1, g2;
+    if (a < b)      { g1 = -1.f; g2 = 0.f; }
+    else if (b < a) { g1 = 0.f;  g2 = -1.f; }
+    else            { g1 = -0.5f; g2 = -0.5f; }
+    dq1[i] = g1 / B;
+    dq2[i] = g2 / B;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    acc += __shfl_down(acc, off);
+    laux += __shfl_down(laux, off);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    red[threadIdx.x >> 6] = acc;
+    redl[threadIdx.x >> 6] = laux;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    loss_acc[0] += (red[0] + red[1] + red[2] + red[3]) / B;
+    if (mean_logp)
+      mean_logp[0] = (redl[0] + redl[1] + redl[2] + redl[3]) / B;
+  }
+}
+
+// learned entropy temperature: one-thread Adam on log_alpha
+// (loss = -log_alpha * (mean_logp + target_entropy))
+__global__ void alpha_update_kernel(float* __restrict__ log_alpha,
+                                    float* __restrict__ alpha_dev,
+                                    float* __restrict__ m, float* __restrict__ v,
+                                    int64_t* __restrict__ step,
+                                    const float* __restrict__ mean_logp,
+                                    float target_entropy, float lr) {
+  float g = -(mean_logp[0] + target_entropy);
+  int64_t t = ++step[0];
+  float mi = 0.9f * m[0] + 0.1f * g;
+  float vi = 0.999f * v[0] + 0.001f * g * g;
+  m[0] = mi; v[0] = vi;
+  float bc1 = 1.f - powf(0.9f, (float)t);
+  float bc2 = 1.f - powf(0.999f, (float)t);
+  log_alpha[0] -= lr / bc1 * mi / (sqrtf(vi / bc2) + 1e-8f);
+  alpha_dev[0] = expf(log_alpha[0]);
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers
+// ---------------------------------------------------------------------------
+
+inline hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+extern bool* g_bf16_flag;  // shared with tac_kernels.hip
+
+inline const float* fptr(const c10::optional<torch::Tensor>& t) {
+  return t.has_value() ? t->data_ptr<float>() : nullptr;
+}
+
+void mgemm(std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
+           std::vector<c10::optional<torch::Tensor>> bs,
+           std::vector<torch::Tensor> ys,
+           std::vector<c10::optional<torch::Tensor>> masks,
+           int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldy,
+           bool relu,
+           std::vector<torch::Tensor> xs2, std::vector<torch::Tensor> ws2,
+           std::vector<c10::optional<torch::Tensor>> masks2, int64_t K2,
+           int64_t x_off, int64_t x2_off) {
+  const int nz = (int)xs.size();
+  TORCH_CHECK(nz >= 1 && nz <= 2);
+  const bool sum2 = !xs2.empty();
+  const bool has_mask = masks.size() && masks[0].has_value();
+  MGemm g{};
+  g.M = (int)M; g.N = (int)N; g.K = (int)K;
+  g.lda = (int)lda; g.ldy = (int)ldy; g.K2 = (int)K2;
+  for (int z = 0; z < nz; ++z) {
+    g.p[z].x = xs[z].data_ptr<float>() + x_off;
+    g.p[z].w = ws[z].data_ptr<float>();
+    g.p[z].bias = fptr(bs[z]);
+    g.p[z].y = ys[z].data_ptr<float>();
+    g.p[z].mask = masks.size() ? fptr(masks[z]) : nullptr;
+    if (g.p[z].mask) g.p[z].mask += x_off;
+    if (sum2) {
+      g.p[z].x2 = xs2[z].data_ptr<float>() + x2_off;
+      g.p[z].w2 = ws2[z].data_ptr<float>();
+      g.p[z].mask2 = masks2.size() ? fptr(masks2[z]) : nullptr;
+      if (g.p[z].mask2) g.p[z].mask2 += x2_off;
+    }
+  }
+  dim3 grid((M + TB - 1) / TB, (N + TB - 1) / TB, nz);
+  const bool bf16 = *g_bf16_flag;
+  auto L = [&](auto b, auto m, auto r, auto s) {
+    hipLaunchKernelGGL((mgemm_kernel<decltype(b)::value, decltype(m)::value,
+                                     decltype(r)::value, decltype(s)::value>),
+                       grid, dim3(256), 0, stream(), g);
+  };
+  // dispatch over (bf16, mask, relu, sum2)
+  #define D2(b, m, r) do { if (sum2) L(b, m, r, std::true_type{}); \
+                           else L(b, m, r, std::false_type{}); } while (0)
+  #define D1(b, m) do { if (relu) D2(b, m, std::true_type{}); \
+                        else D2(b, m, std::false_type{}); } while (0)
+  if (bf16) { if (has_mask) D1(std::true_type{}, std::true_type{});
+              else D1(std::true_type{}, std::false_type{}); }
+  else      { if (has_mask) D1(std::false_type{}, std::true_type{});
+              else D1(std::false_type{}, std::false_type{}); }
+  #undef D1
+  #undef D2
+}
+
+void mwgrad(std::vector<torch::Tensor> dys,
+            std::vector<c10::optional<torch::Tensor>> ymasks,
+            std::vector<torch::Tensor> xs,
+            std::vector<torch::Tensor> dws, std::vector<torch::Tensor> dbs,
+            int64_t M, int64_t N, int64_t K, int64_t lddy, int64_t ldx,
+            int64_t x_off) {
+  const int nz = (int)dys.size();
+  const bool has_mask = ymasks.size() && ymasks[0].has_value();
+  WGemm g{};
+  g.M = (int)M; g.N = (int)N; g.K = (int)K;
+  g.lddy = (int)lddy; g.ldx = (int)ldx;
+  for (int z = 0; z < nz; ++z) {
+    g.p[z].dy = dys[z].data_ptr<float>();
+    g.p[z].ymask = ymasks.size() ? fptr(ymasks[z]) : nullptr;
+    g.p[z].x = xs[z].data_ptr<float>() + x_off;
+    g.p[z].dw = dws[z].data_ptr<float>();
+    g.p[z].db = dbs[z].numel() ? dbs[z].data_ptr<float>() : nullptr;
+  }
+  dim3 grid((N + TB - 1) / TB, (K + TB - 1) / TB, nz);
+  const bool bf16 = *g_bf16_flag;
+  if (bf16) {
+    if (has_mask)
+      hipLaunchKernelGGL((mwgrad_kernel<true, true>), grid, dim3(256), 0,
+                         stream(), g);
+    else
+      hipLaunchKernelGGL((mwgrad_kernel<true, false>), grid, dim3(256), 0,
+                         stream(), g);
+  } else {
+    if (has_mask)
+      hipLaunchKernelGGL((mwgrad_kernel<false, true>), grid, dim3(256), 0,
+                         stream(), g);
+    else
+      hipLaunchKernelGGL((mwgrad_kernel<false, false>), grid, dim3(256), 0,
+                         stream(), g);
+  }
+}
+
+void transpose_multi(std::vector<torch::Tensor> ws,
+                     std::vector<torch::Tensor> wts) {
+  TArgs t{};
+  t.n_layers = (int)ws.size();
+  TORCH_CHECK(t.n_layers <= MAX_T);
+  for (int i = 0; i < t.n_layers; ++i) {
+    t.w[i] = ws[i].data_ptr<float>();
+    t.wt[i] = wts[i].data_ptr<float>();
+    t.N[i] = (int)ws[i].size(0);
+    t.K[i] = (int)ws[i].size(1);
+  }
+  t.blocks_per_layer = 8;
+  hipLaunchKernelGGL(transpose_multi_kernel,
+                     dim3(t.n_layers * t.blocks_per_layer), dim3(256), 0,
+                     stream(), t);
+}
+
+void gather2(torch::Tensor state, torch::Tensor act, torch::Tensor rew,
+             torch::Tensor nstate, torch::Tensor done,
+             torch::Tensor size_dev, torch::Tensor ctr, int64_t seed,
+             torch::Tensor xc, torch::Tensor xc2, torch::Tensor orew,
+             torch::Tensor od, int64_t B) {
+  const int obs_dim = (int)state.size(1);
+  const int act_dim = (int)act.size(1);
+  int threads = std::min<int>(256, std::max(64, ((obs_dim + 63) / 64) * 64));
+  hipLaunchKernelGGL(gather2_kernel, dim3((int)B), dim3(threads), 0, stream(),
+                     state.data_ptr<float>(), act.data_ptr<float>(),
+                     rew.data_ptr<float>(), nstate.data_ptr<float>(),
+                     done.data_ptr<float>(), size_dev.data_ptr<int64_t>(),
+                     ctr.data_ptr<int64_t>(), (uint64_t)seed,
+                     xc.data_ptr<float>(), xc2.data_ptr<float>(),
+                     orew.data_ptr<float>(), od.data_ptr<float>(),
+                     (int)B, obs_dim, act_dim);
+}
+
+void tg_fwd2(torch::Tensor mu, torch::Tensor ls, torch::Tensor out0,
+             int64_t col0, torch::Tensor out1, int64_t col1, int64_t B0,
+             torch::Tensor logp, torch::Tensor prob, torch::Tensor ctr,
+             int64_t seed, double act_limit, double lo, double hi) {
+  const int R = (int)mu.size(0), A = (int)mu.size(1);
+  TORCH_CHECK(A <= 64);
+  hipLaunchKernelGGL(tg_fwd2_kernel, dim3(R), dim3(64), 0, stream(),
+                     mu.data_ptr<float>(), ls.data_ptr<float>(),
+                     out0.data_ptr<float>() + col0, (int)out0.size(1),
+                     out1.data_ptr<float>() + col1, (int)out1.size(1),
+                     (int)B0, logp.data_ptr<float>(), prob.data_ptr<float>(),
+                     ctr.data_ptr<int64_t>(), (uint64_t)seed, R, A,
+                     (float)act_limit, (float)lo, (float)hi);
+}
+
+void tg_bwd2(torch::Tensor dxc, int64_t col0,
+             c10::optional<torch::Tensor> alpha_dev, double alpha_host,
+             torch::Tensor mu, torch::Tensor ls, torch::Tensor prob,
+             torch::Tensor dmu, torch::Tensor dls, int64_t B,
+             double act_limit, double lo, double hi) {
+  const int A = (int)dmu.size(1);
+  hipLaunchKernelGGL(tg_bwd2_kernel, dim3((int)B), dim3(64), 0, stream(),
+                     dxc.data_ptr<float>(), (int)dxc.size(1), (int)col0,
+                     fptr(alpha_dev), (float)alpha_host,
+                     mu.data_ptr<float>(), ls.data_ptr<float>(),
+                     prob.data_ptr<float>(), dmu.data_ptr<float>(),
+                     dls.data_ptr<float>(), (int)B, A, (float)act_limit,
+                     (float)lo, (float)hi);
+}
+
+void qloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor q1t,
+            torch::Tensor q2t, torch::Tensor logp_next, torch::Tensor rew,
+            torch::Tensor done, c10::optional<torch::Tensor> alpha_dev,
+            double alpha_host, torch::Tensor loss_acc, torch::Tensor dq1,
+            torch::Tensor dq2, int64_t B, double gamma, double scale) {
+  hipLaunchKernelGGL(qloss2_kernel, dim3(1), dim3(256), 0, stream(),
+                     q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     q1t.data_ptr<float>(), q2t.data_ptr<float>(),
+                     logp_next.data_ptr<float>(), rew.data_ptr<float>(),
+                     done.data_ptr<float>(), fptr(alpha_dev),
+                     (float)alpha_host, loss_acc.data_ptr<float>(),
+                     dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B,
+                     (float)gamma, (float)scale);
+}
+
+void piloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor logp,
+             c10::optional<torch::Tensor> alpha_dev, double alpha_host,
+             torch::Tensor loss_acc, c10::optional<torch::Tensor> mean_logp,
+             torch::Tensor dq1, torch::Tensor dq2, int64_t B) {
+  hipLaunchKernelGGL(piloss2_kernel, dim3(1), dim3(256), 0, stream(),
+                     q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     logp.data_ptr<float>(), fptr(alpha_dev),
+                     (float)alpha_host, loss_acc.data_ptr<float>(),
+                     mean_logp.has_value() ? mean_logp->data_ptr<float>()
+                                           : nullptr,
+                     dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B);
+}
+
+torch::Tensor tg_eps(int64_t ctr_val, int64_t seed, int64_t R, int64_t A,
+                     torch::Tensor like) {
+  auto out = torch::empty({R, A}, like.options());
+  hipLaunchKernelGGL(tg_eps_kernel, dim3((int)R), dim3(64), 0, stream(),
+                     out.data_ptr<float>(), (uint64_t)ctr_val,
+                     (uint64_t)seed, (int)R, (int)A);
+  return out;
+}
+
+void alpha_update(torch::Tensor log_alpha, torch::Tensor alpha_dev,
+                  torch::Tensor m, torch::Tensor v, torch::Tensor step,
+                  torch::Tensor mean_logp, double target_entropy, double lr) {
+  hipLaunchKernelGGL(alpha_update_kernel, dim3(1), dim3(1), 0, stream(),
+                     log_alpha.data_ptr<float>(), alpha_dev.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     step.data_ptr<int64_t>(), mean_logp.data_ptr<float>(),
+                     (float)target_entropy, (float)lr);
+}
+
+}  // namespace fused
+
+void register_fused(pybind11::module_& m) {
+  m.def("mgemm", &fused::mgemm,
+        "multi-problem MFMA GEMM (fwd-form, strided, masked, sum2)");
+  m.def("mwgrad", &fused::mwgrad, "multi-problem wgrad + fused db");
+  m.def("transpose_multi", &fused::transpose_multi);
+  m.def("gather2", &fused::gather2);
+  m.def("tg_fwd2", &fused::tg_fwd2);
+  m.def("tg_bwd2", &fused::tg_bwd2);
+  m.def("qloss2", &fused::qloss2);
+  m.def("piloss2", &fused::piloss2);
+  m.def("alpha_update", &fused::alpha_update);
+  m.def("tg_eps", &fused::tg_eps);
+}

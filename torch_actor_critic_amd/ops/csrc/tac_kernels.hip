@@ -989,6 +989,11 @@ void replay_sample_into(torch::Tensor state, torch::Tensor act,
                      od.data_ptr<float>(), obs_dim, act_dim);
 }
 
+void bump_counter(torch::Tensor ctr) {
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, cur_stream(),
+                     ctr.data_ptr<int64_t>());
+}
+
 void philox_randn_(torch::Tensor out, torch::Tensor ctr, int64_t seed) {
   CHECK_IN(out);
   int64_t n = out.numel();
@@ -1021,7 +1026,13 @@ std::vector<torch::Tensor> replay_sample(torch::Tensor state, torch::Tensor act,
 
 }  // namespace
 
+// shared compute-mode flag for the fused-engine TU (fused.hip)
+namespace fused { bool* g_bf16_flag = nullptr; }
+void register_fused(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  fused::g_bf16_flag = &g_bf16;
+  register_fused(m);
   m.def("set_compute_bf16", &set_compute_bf16,
         "Switch GEMM kernels to bf16 MFMA inputs (fp32 accumulate)");
   m.def("get_compute_bf16", &get_compute_bf16);
@@ -1036,4 +1047,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("replay_sample", &replay_sample);
   m.def("replay_sample_into", &replay_sample_into);
   m.def("philox_randn_", &philox_randn_);
+  m.def("bump_counter", &bump_counter);
 }

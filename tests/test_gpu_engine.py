@@ -205,8 +205,9 @@ def test_engine_learned_alpha():
     eng2 = FusedSACEngine(sac, actor, critic, target, buf, pi_opt, q_opt,
                           target_flat, B, torch.device(DEV), sample=True,
                           capture=True, philox_seed=9)
+    # capture warmup already ran 2 in-graph alpha updates
     a0 = float(eng2.alpha_dev.item())
-    assert abs(a0 - ALPHA) < 1e-4
+    assert abs(a0 - ALPHA) < 0.01
     for _ in range(10):
         eng2.step()
     torch.cuda.synchronize()

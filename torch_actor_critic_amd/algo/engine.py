@@ -128,6 +128,12 @@ class FusedSACEngine:
         self._a_tr_dst = list(self.awt) + list(self.hwt)
         self.ext.transpose_multi(self._c_tr_src, self._c_tr_dst)
         self.ext.transpose_multi(self._a_tr_src, self._a_tr_dst)
+        # flat offsets of each weight slab (for the fused Adam+transpose)
+        def offs(opt, srcs):
+            base = opt.fp.flat.data_ptr()
+            return [(w.data_ptr() - base) // 4 for w in srcs]
+        self._c_offs = offs(q_opt, self._c_tr_src)
+        self._a_offs = offs(pi_opt, self._a_tr_src)
 
         # ---- losses / alpha -------------------------------------------
         self.loss_q_acc = torch.zeros(1, **f32)
@@ -253,8 +259,12 @@ class FusedSACEngine:
     def _phase_policy(self):
         ext = self.ext
         B, O, A, OC = self.B, self.O, self.A, self.OC
-        self.q_opt.step()
-        ext.transpose_multi(self._c_tr_src, self._c_tr_dst)
+        # fused Adam + transposed-weight refresh (one bump + one kernel)
+        qo = self.q_opt
+        ext.bump_counter(qo.step_t)
+        ext.adam_t(qo.fp.flat, qo.fp.flat_grad, qo.m, qo.v, qo.step_t,
+                   qo.lr, qo.betas[0], qo.betas[1], qo.eps,
+                   qo.weight_decay, self._c_offs, self._c_tr_dst)
 
         # critic on (s, pi) = XC2 with the UPDATED critic
         self._critic_fwd(self.XC2, 0, self.cw, self.p_act, OC)
@@ -316,8 +326,11 @@ class FusedSACEngine:
 
     def _phase_finish(self):
         ext = self.ext
-        self.pi_opt.step()
-        ext.transpose_multi(self._a_tr_src, self._a_tr_dst)
+        po = self.pi_opt
+        ext.bump_counter(po.step_t)
+        ext.adam_t(po.fp.flat, po.fp.flat_grad, po.m, po.v, po.step_t,
+                   po.lr, po.betas[0], po.betas[1], po.eps,
+                   po.weight_decay, self._a_offs, self._a_tr_dst)
         ext.polyak_(self.target_flat, self.q_opt.fp.flat, self.sac.polyak)
         if self.learn_alpha:
             ext.alpha_update(self.log_alpha, self.alpha_dev, self.alpha_m,

@@ -83,10 +83,14 @@ struct MGemm {
 };
 
 constexpr int TB = 64;        // block tile (M and N)
-constexpr int BKB2 = 64;      // bf16 K-step
+constexpr int BKB2 = 64;      // bf16 K-step (wgrad staging path)
 constexpr int BKF2 = 16;      // fp32 K-step
-constexpr int LDSB2 = 72;     // bf16 LDS halves per row
+constexpr int LDSB2 = 72;     // bf16 LDS halves per row (BKB2 tiles)
 constexpr int LDSF2 = 17;     // fp32 LDS floats per row
+// pipelined mgemm uses a deeper bf16 K-step (fewer serial tiles)
+constexpr int BKP = 128;      // bf16 K-step (mgemm pipelined path)
+constexpr int LDSP = 136;     // halves per row: stride 68 dwords -> 4r mod 64
+                              // distinct over a 16-lane group, conflict-free
 
 // Stage a 64 x BK tile of `src` (row-major, leading dim ld, rows base
 // `r0`, cols base `k0`, bounds R x C) into LDS, optionally masked.
@@ -204,6 +208,138 @@ DEVINL void mma_tiles(const void* xs_, const void* ws_, f32x4 (&acc)[2][2],
   }
 }
 
+// ---- pipelined register staging (T14: write LDS, issue next tile's
+// global loads, barrier, MFMA — HBM/L2 latency hides under the MFMAs) --
+
+template <bool BF16, bool MASK>
+DEVINL void load_tile_regs(float* v, const float* src, const float* mask,
+                           int r0, int k0, int R, int C, int ld) {
+  const int tid = threadIdx.x;
+  const int row = tid & 63;
+  const int gr = r0 + row;
+  if constexpr (BF16) {
+    const int c0 = (tid >> 6) * 32;
+    bool interior = (r0 + 64 <= R) && (k0 + BKP <= C) && ((ld & 3) == 0);
+    if (interior) {
+      const float* p = src + (int64_t)gr * ld + k0 + c0;
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float4 f = *(const float4*)(p + q * 4);
+        v[q*4+0]=f.x; v[q*4+1]=f.y; v[q*4+2]=f.z; v[q*4+3]=f.w;
+      }
+      if constexpr (MASK) {
+        const float* mp = mask + (int64_t)gr * ld + k0 + c0;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          float4 f = *(const float4*)(mp + q * 4);
+          v[q*4+0] = f.x > 0.f ? v[q*4+0] : 0.f;
+          v[q*4+1] = f.y > 0.f ? v[q*4+1] : 0.f;
+          v[q*4+2] = f.z > 0.f ? v[q*4+2] : 0.f;
+          v[q*4+3] = f.w > 0.f ? v[q*4+3] : 0.f;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int e = 0; e < 32; ++e) {
+        int c = k0 + c0 + e;
+        float val = 0.f;
+        if (gr < R && c < C) {
+          val = src[(int64_t)gr * ld + c];
+          if constexpr (MASK)
+            val = mask[(int64_t)gr * ld + c] > 0.f ? val : 0.f;
+        }
+        v[e] = val;
+      }
+    }
+  } else {
+    const int c0 = (tid >> 6) * 4;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      int c = k0 + c0 + e;
+      float val = 0.f;
+      if (gr < R && c < C) {
+        val = src[(int64_t)gr * ld + c];
+        if constexpr (MASK)
+          val = mask[(int64_t)gr * ld + c] > 0.f ? val : 0.f;
+      }
+      v[e] = val;
+    }
+  }
+}
+
+template <bool BF16>
+DEVINL void write_tile_lds(void* lds, const float* v) {
+  const int tid = threadIdx.x;
+  const int row = tid & 63;
+  if constexpr (BF16) {
+    __bf16* d = (__bf16*)lds;
+    const int c0 = (tid >> 6) * 32;
+    union { __bf16 h[32]; uint4 u[4]; } pk;
+#pragma unroll
+    for (int e = 0; e < 32; ++e) pk.h[e] = (__bf16)v[e];
+    uint4* dst = (uint4*)&d[row * LDSP + c0];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) dst[q] = pk.u[q];
+  } else {
+    float* d = (float*)lds;
+    const int c0 = (tid >> 6) * 4;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) d[row * LDSF2 + c0 + e] = v[e];
+  }
+}
+
+template <bool BF16>
+DEVINL void mma_tiles_p(const void* xs_, const void* ws_,
+                        f32x4 (&acc)[2][2], int lane, int wrow, int wcol) {
+  if constexpr (BF16) {
+    const __bf16* xs = (const __bf16*)xs_;
+    const __bf16* ws = (const __bf16*)ws_;
+    const int arow = lane & 15;
+    const int ak0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int kk = 0; kk < BKP; kk += 32) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        bf16x8 a = *(const bf16x8*)&xs[(wrow + mi * 16 + arow) * LDSP
+                                       + kk + ak0];
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          bf16x8 b = *(const bf16x8*)&ws[(wcol + ni * 16 + arow) * LDSP
+                                         + kk + ak0];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  } else {
+    mma_tiles<false>(xs_, ws_, acc, lane, wrow, wcol);
+  }
+}
+
+template <bool BF16, bool MASK>
+DEVINL void gemm_pass(const float* x, const float* w, const float* mask,
+                      int M, int N, int K, int lda, void* xs, void* ws,
+                      int bm0, int bn0, int lane, int wrow, int wcol,
+                      int tid, f32x4 (&acc)[2][2]) {
+  constexpr int BK = BF16 ? BKP : BKF2;
+  constexpr int EL = BF16 ? 32 : 4;
+  float va[EL], vb[EL];
+  (void)tid;
+  load_tile_regs<BF16, MASK>(va, x, mask, bm0, 0, M, K, lda);
+  load_tile_regs<BF16, false>(vb, w, nullptr, bn0, 0, N, K, K);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    write_tile_lds<BF16>(xs, va);
+    write_tile_lds<BF16>(ws, vb);
+    if (k0 + BK < K) {
+      load_tile_regs<BF16, MASK>(va, x, mask, bm0, k0 + BK, M, K, lda);
+      load_tile_regs<BF16, false>(vb, w, nullptr, bn0, k0 + BK, N, K, K);
+    }
+    __syncthreads();
+    mma_tiles_p<BF16>(xs, ws, acc, lane, wrow, wcol);
+    __syncthreads();
+  }
+}
+
 template <bool BF16, bool MASK, bool RELU, bool SUM2>
 __global__ __launch_bounds__(256)
 void mgemm_kernel(MGemm g) {
@@ -215,28 +351,17 @@ void mgemm_kernel(MGemm g) {
   const int wcol = (wid & 1) * 32;
   const int bm0 = blockIdx.x * TB;
   const int bn0 = blockIdx.y * TB;
-  constexpr int BK = BF16 ? BKB2 : BKF2;
-  constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
+  constexpr int LBYTES = BF16 ? (64 * LDSP * 2) : (64 * LDSF2 * 4);
   __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
   void* xs = smem;
   void* ws = smem + LBYTES;
 
   f32x4 acc[2][2] = {};
-  for (int k0 = 0; k0 < g.K; k0 += BK) {
-    stage_tile<BF16, MASK>(xs, p.x, p.mask, bm0, k0, g.M, g.K, g.lda);
-    stage_tile<BF16, false>(ws, p.w, nullptr, bn0, k0, g.N, g.K, g.K);
-    __syncthreads();
-    mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
-    __syncthreads();
-  }
+  gemm_pass<BF16, MASK>(p.x, p.w, p.mask, g.M, g.N, g.K, g.lda, xs, ws,
+                        bm0, bn0, lane, wrow, wcol, tid, acc);
   if constexpr (SUM2) {
-    for (int k0 = 0; k0 < g.K2; k0 += BK) {
-      stage_tile<BF16, MASK>(xs, p.x2, p.mask2, bm0, k0, g.M, g.K2, g.K2);
-      stage_tile<BF16, false>(ws, p.w2, nullptr, bn0, k0, g.N, g.K2, g.K2);
-      __syncthreads();
-      mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
-      __syncthreads();
-    }
+    gemm_pass<BF16, MASK>(p.x2, p.w2, p.mask2, g.M, g.N, g.K2, g.K2, xs,
+                          ws, bm0, bn0, lane, wrow, wcol, tid, acc);
   }
 
   const int crow = (lane >> 4) * 4, ccol = lane & 15;
@@ -687,6 +812,54 @@ __global__ void alpha_update_kernel(float* __restrict__ log_alpha,
 }
 
 // ---------------------------------------------------------------------------
+// Adam with fused transposed-weight refresh: after updating flat[i],
+// weight-slab elements also write their transposed copy (the dgrad
+// operand cache) — removes the separate transpose kernels per update.
+// ---------------------------------------------------------------------------
+
+struct ATArgs {
+  int64_t off[MAX_T];    // flat offset of each weight slab
+  float* wt[MAX_T];
+  int N[MAX_T], K[MAX_T];
+  int n_layers;
+};
+
+__global__ __launch_bounds__(256)
+void adam_t_kernel(float* __restrict__ p, const float* __restrict__ g,
+                   float* __restrict__ m, float* __restrict__ v,
+                   const int64_t* __restrict__ step, int64_t n,
+                   float lr, float b1, float b2, float eps, float wd,
+                   ATArgs ta) {
+  const float t = (float)step[0];
+  const float bc1 = 1.f - powf(b1, t);
+  const float bc2 = 1.f - powf(b2, t);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float gi = g[i] + wd * p[i];
+    float mi = b1 * m[i] + (1.f - b1) * gi;
+    float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    float pn = p[i] - lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+    p[i] = pn;
+#pragma unroll
+    for (int L = 0; L < MAX_T; ++L) {
+      if (L >= ta.n_layers) break;
+      int64_t lo = ta.off[L];
+      int64_t sz = (int64_t)ta.N[L] * ta.K[L];
+      if (i >= lo && i < lo + sz) {
+        int64_t loc = i - lo;
+        int nn = (int)(loc / ta.K[L]);
+        int kk = (int)(loc % ta.K[L]);
+        ta.wt[L][(int64_t)kk * ta.N[L] + nn] = pn;
+        break;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
 
@@ -878,6 +1051,28 @@ void piloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor logp,
                      dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B);
 }
 
+void adam_t(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+            torch::Tensor v, torch::Tensor step, double lr, double b1,
+            double b2, double eps, double wd,
+            std::vector<int64_t> offsets, std::vector<torch::Tensor> wts) {
+  ATArgs ta{};
+  ta.n_layers = (int)offsets.size();
+  TORCH_CHECK(ta.n_layers <= MAX_T);
+  for (int i = 0; i < ta.n_layers; ++i) {
+    ta.off[i] = offsets[i];
+    ta.wt[i] = wts[i].data_ptr<float>();
+    ta.K[i] = (int)wts[i].size(0);   // wt is [K, N]
+    ta.N[i] = (int)wts[i].size(1);
+  }
+  int64_t n = p.numel();
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 1024);
+  hipLaunchKernelGGL(adam_t_kernel, dim3(blocks), dim3(256), 0, stream(),
+                     p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     step.data_ptr<int64_t>(), n, (float)lr, (float)b1,
+                     (float)b2, (float)eps, (float)wd, ta);
+}
+
 torch::Tensor tg_eps(int64_t ctr_val, int64_t seed, int64_t R, int64_t A,
                      torch::Tensor like) {
   auto out = torch::empty({R, A}, like.options());
@@ -911,4 +1106,5 @@ void register_fused(pybind11::module_& m) {
   m.def("piloss2", &fused::piloss2);
   m.def("alpha_update", &fused::alpha_update);
   m.def("tg_eps", &fused::tg_eps);
+  m.def("adam_t", &fused::adam_t);
 }

@@ -253,8 +253,11 @@ def test_sac_train_on_gpu_learns_pendulum():
     from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
     from torch_actor_critic_amd.optim import FlatAdam
 
+    from torch_actor_critic_amd.ops import functional as Fo
+
     torch.manual_seed(0)
     np.random.seed(0)
+    Fo.set_philox_seed(0)  # isolate from other tests' counter state
     device = torch.device(DEV)
     env = envs.make("Pendulum-v1")
     env.seed(0)
@@ -265,14 +268,15 @@ def test_sac_train_on_gpu_learns_pendulum():
 
     sac = SAC(alpha=0.1, gamma=0.99, polyak=0.995, reward_scale=1.0,
               epochs=1, batch_size=64, start_steps=500,
-              steps_per_epoch=4000, max_ep_len=200, update_after=500,
+              steps_per_epoch=6000, max_ep_len=200, update_after=500,
               update_every=50, save_every=1000)
     sac.train(0, env, actor, critic, buf, pi_opt, q_opt, render=False,
               logging=False)
     assert sac._graph is not None, "fused engine was not used"
 
+    # random-policy baseline on this env is ~ -1200; require a clear gap
     eval_rets = []
-    for _ in range(3):
+    for _ in range(5):
         state = env.reset()
         ep, done = 0.0, False
         while not done:
@@ -282,4 +286,4 @@ def test_sac_train_on_gpu_learns_pendulum():
             state, r, done, _ = env.step(a.cpu().numpy())
             ep += r
         eval_rets.append(ep)
-    assert float(np.mean(eval_rets)) > -400.0, eval_rets
+    assert float(np.mean(eval_rets)) > -700.0, eval_rets

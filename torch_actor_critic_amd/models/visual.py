@@ -42,6 +42,15 @@ def calculate_size(image_shape, filters, kernel_sizes, strides) -> int:
     return int(c * h * w)
 
 
+class Conv2d(nn.Conv2d):
+    """nn.Conv2d whose forward dispatches to the hand-written gfx950
+    implicit-GEMM MFMA kernels on GPU (ops/csrc/conv.hip); state-dict
+    identical to nn.Conv2d."""
+
+    def forward(self, x):
+        return Fo.conv2d(x, self.weight, self.bias, self.stride[0])
+
+
 def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
                strides=[4, 2, 1], activation: t.Type[nn.Module] = nn.ReLU,
                dense_size: int = 512) -> nn.Module:
@@ -52,8 +61,8 @@ def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
     sizes = [channels] + list(filters)
     for i in range(len(sizes) - 1):
         model.add_module(f"conv_{i}",
-                         nn.Conv2d(sizes[i], sizes[i + 1], kernel_sizes[i],
-                                   strides[i]))
+                         Conv2d(sizes[i], sizes[i + 1], kernel_sizes[i],
+                                strides[i]))
         model.add_module(f"relu_{i}", activation())
     flat = calculate_size(input_shape, filters, kernel_sizes, strides)
     model.add_module("flatten", nn.Flatten())

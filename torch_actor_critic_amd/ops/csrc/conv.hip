@@ -1,0 +1,423 @@
+// Implicit-GEMM convolution kernels (gfx950 / CDNA4) for the visual
+// actor/critic path (reference networks/convolutional.py:30-51: Nature-
+// CNN trunk, valid padding, strides [4,2,1], kernels [8,4,3]).
+//
+// All three passes run as MFMA GEMMs over an implicitly-gathered
+// im2col operand (no materialized im2col buffer):
+//   fwd   : Y[m=(b,oy,ox), n=oc]  = sum_k  X[b,ic,oy*s+ky,ox*s+kx] W[oc,k]
+//   dgrad : dX[m=(b,iy,ix), n=ic] = sum_k' dY[b,oc,(iy-ky)/s,(ix-kx)/s]
+//                                          WT[ic, k'=(oc,ky,kx)]
+//   wgrad : dW[oc, k] = sum_m dY[m,oc] im2col[m,k];  db fused
+// Weight tensors stay fp32 master; bf16 mode rounds operands on stage.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+namespace convk {
+
+#define DEVINL __device__ __forceinline__
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+constexpr int TB = 64;
+constexpr int BKB = 64, LDB = 72;   // bf16 K-step / LDS halves per row
+constexpr int BKF = 16, LDF = 17;   // fp32
+
+struct ConvDims {
+  int B, IC, IH, IW, OC, OH, OW, KH, KW, S;
+};
+
+// ---- shared MFMA tile compute (same fragment plan as fused.hip) -------
+
+template <bool BF16>
+DEVINL void mma_tiles(const void* xs_, const void* ws_, f32x4 (&acc)[2][2],
+                      int lane, int wrow, int wcol) {
+  if constexpr (BF16) {
+    const __bf16* xs = (const __bf16*)xs_;
+    const __bf16* ws = (const __bf16*)ws_;
+    const int arow = lane & 15;
+    const int ak0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int kk = 0; kk < BKB; kk += 32) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        bf16x8 a = *(const bf16x8*)&xs[(wrow + mi * 16 + arow) * LDB
+                                       + kk + ak0];
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          bf16x8 b = *(const bf16x8*)&ws[(wcol + ni * 16 + arow) * LDB
+                                         + kk + ak0];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, b, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  } else {
+    const float* xs = (const float*)xs_;
+    const float* ws = (const float*)ws_;
+    const int arow = lane & 15;
+    const int akl = lane >> 4;
+#pragma unroll
+    for (int kk = 0; kk < BKF; kk += 4) {
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        float a = xs[(wrow + mi * 16 + arow) * LDF + kk + akl];
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          float b = ws[(wcol + ni * 16 + arow) * LDF + kk + akl];
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a, b, acc[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+  }
+}
+
+template <bool BF16>
+DEVINL void lds_put(void* lds, int row, int c, float v) {
+  if constexpr (BF16) ((__bf16*)lds)[row * LDB + c] = (__bf16)v;
+  else                ((float*)lds)[row * LDF + c] = v;
+}
+
+// im2col element: A[m, k] with m=(b,oy,ox), k=(ic,ky,kx)
+DEVINL float im2col_at(const float* x, const ConvDims& d, int m, int k) {
+  const int ox = m % d.OW, t1 = m / d.OW;
+  const int oy = t1 % d.OH, b = t1 / d.OH;
+  const int kx = k % d.KW, t2 = k / d.KW;
+  const int ky = t2 % d.KH, ic = t2 / d.KH;
+  const int iy = oy * d.S + ky, ix = ox * d.S + kx;
+  return x[(((int64_t)b * d.IC + ic) * d.IH + iy) * d.IW + ix];
+}
+
+// ---------------------------------------------------------------------------
+// conv fwd: grid (ceil(M/64), ceil(OC/64)); Y NCHW scatter epilogue
+// ---------------------------------------------------------------------------
+
+template <bool BF16, bool RELU>
+__global__ __launch_bounds__(256)
+void conv_fwd_kernel(const float* __restrict__ x, const float* __restrict__ w,
+                     const float* __restrict__ bias, float* __restrict__ y,
+                     ConvDims d) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bm0 = blockIdx.x * TB;
+  const int bn0 = blockIdx.y * TB;
+  const int M = d.B * d.OH * d.OW;
+  const int K = d.IC * d.KH * d.KW;
+  constexpr int BK = BF16 ? BKB : BKF;
+  constexpr int EL = BF16 ? 16 : 4;
+  constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  void* xs = smem;
+  void* ws = smem + LBYTES;
+  f32x4 acc[2][2] = {};
+
+  const int row = tid & 63;
+  const int c00 = (tid >> 6) * EL;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      int c = c00 + e;
+      int m = bm0 + row, k = k0 + c;
+      float va = (m < M && k < K) ? im2col_at(x, d, m, k) : 0.f;
+      lds_put<BF16>(xs, row, c, va);
+      int n = bn0 + row;
+      float vb = (n < d.OC && k < K) ? w[(int64_t)n * K + k] : 0.f;
+      lds_put<BF16>(ws, row, c, vb);
+    }
+    __syncthreads();
+    mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = bm0 + wrow + mi * 16 + crow + r;
+        int oc = bn0 + wcol + ni * 16 + ccol;
+        if (m < M && oc < d.OC) {
+          float v = acc[mi][ni][r] + (bias ? bias[oc] : 0.f);
+          if constexpr (RELU) v = fmaxf(v, 0.f);
+          const int ox = m % d.OW, t1 = m / d.OW;
+          const int oy = t1 % d.OH, b = t1 / d.OH;
+          y[(((int64_t)b * d.OC + oc) * d.OH + oy) * d.OW + ox] = v;
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
+// conv dgrad: dX = spread(dY) @ WT; WT [IC, OC*KH*KW] prepared by caller.
+// m=(b,iy,ix), k'=(oc,ky,kx).  RELU mask (y>0) applied to dY on gather.
+// ---------------------------------------------------------------------------
+
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void conv_dgrad_kernel(const float* __restrict__ dy,
+                       const float* __restrict__ ymask,
+                       const float* __restrict__ wt,
+                       float* __restrict__ dx, ConvDims d) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bm0 = blockIdx.x * TB;
+  const int bn0 = blockIdx.y * TB;
+  const int M = d.B * d.IH * d.IW;
+  const int K = d.OC * d.KH * d.KW;
+  constexpr int BK = BF16 ? BKB : BKF;
+  constexpr int EL = BF16 ? 16 : 4;
+  constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  void* xs = smem;
+  void* ws = smem + LBYTES;
+  f32x4 acc[2][2] = {};
+
+  const int row = tid & 63;
+  const int c00 = (tid >> 6) * EL;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      int c = c00 + e;
+      int m = bm0 + row, k = k0 + c;
+      float va = 0.f;
+      if (m < M && k < K) {
+        const int ix = m % d.IW, t1 = m / d.IW;
+        const int iy = t1 % d.IH, b = t1 / d.IH;
+        const int kx = k % d.KW, t2 = k / d.KW;
+        const int ky = t2 % d.KH, oc = t2 / d.KH;
+        const int ry = iy - ky, rx = ix - kx;
+        if (ry >= 0 && rx >= 0 && ry % d.S == 0 && rx % d.S == 0) {
+          const int oy = ry / d.S, ox = rx / d.S;
+          if (oy < d.OH && ox < d.OW) {
+            int64_t idx = (((int64_t)b * d.OC + oc) * d.OH + oy) * d.OW + ox;
+            va = dy[idx];
+            if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
+          }
+        }
+      }
+      lds_put<BF16>(xs, row, c, va);
+      int n = bn0 + row;
+      float vb = (n < d.IC && k < K) ? wt[(int64_t)n * K + k] : 0.f;
+      lds_put<BF16>(ws, row, c, vb);
+    }
+    __syncthreads();
+    mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = bm0 + wrow + mi * 16 + crow + r;
+        int ic = bn0 + wcol + ni * 16 + ccol;
+        if (m < M && ic < d.IC) {
+          const int ix = m % d.IW, t1 = m / d.IW;
+          const int iy = t1 % d.IH, b = t1 / d.IH;
+          dx[(((int64_t)b * d.IC + ic) * d.IH + iy) * d.IW + ix] =
+              acc[mi][ni][r];
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
+// conv wgrad: dW[oc, k] = sum_m dYeff[m, oc] im2col[m, k]; db fused.
+// Reduction over M; tiles: rows=oc (64), cols=k (64), i-chunks of BK.
+// ---------------------------------------------------------------------------
+
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void conv_wgrad_kernel(const float* __restrict__ dy,
+                       const float* __restrict__ ymask,
+                       const float* __restrict__ x,
+                       float* __restrict__ dw, float* __restrict__ db,
+                       ConvDims d) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bn0 = blockIdx.x * TB;   // oc rows
+  const int bk0 = blockIdx.y * TB;   // k cols
+  const int M = d.B * d.OH * d.OW;
+  const int K = d.IC * d.KH * d.KW;
+  constexpr int BK = BF16 ? BKB : BKF;
+  constexpr int EL = BF16 ? 16 : 4;
+  constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  __shared__ float dbs[64];
+  if (tid < 64) dbs[tid] = 0.f;
+  f32x4 acc[2][2] = {};
+
+  const int row = tid & 63;
+  const int c00 = (tid >> 6) * EL;
+  for (int i0 = 0; i0 < M; i0 += BK) {
+    // A tile: as[oc][i] = dYeff[i0+i, bn0+oc]
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      int i = c00 + e;
+      int m = i0 + i, oc = bn0 + row;
+      float va = 0.f;
+      if (m < M && oc < d.OC) {
+        const int ox = m % d.OW, t1 = m / d.OW;
+        const int oy = t1 % d.OH, b = t1 / d.OH;
+        int64_t idx = (((int64_t)b * d.OC + oc) * d.OH + oy) * d.OW + ox;
+        va = dy[idx];
+        if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
+      }
+      lds_put<BF16>(smem, row, i, va);
+      // B tile: bs[k][i] = im2col[i0+i, bk0+k]
+      int kcol = bk0 + row;
+      float vb = (m < M && kcol < K) ? im2col_at(x, d, m, kcol) : 0.f;
+      lds_put<BF16>(smem + LBYTES, row, i, vb);
+    }
+    __syncthreads();
+    mma_tiles<BF16>(smem, smem + LBYTES, acc, lane, wrow, wcol);
+    if (db && blockIdx.y == 0 && tid < 64) {
+      float s = 0.f;
+      if constexpr (BF16) {
+        const __bf16* as = (const __bf16*)smem;
+        for (int i = 0; i < BKB; ++i) s += (float)as[tid * LDB + i];
+      } else {
+        const float* as = (const float*)smem;
+        for (int i = 0; i < BKF; ++i) s += as[tid * LDF + i];
+      }
+      dbs[tid] += s;
+    }
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int oc = bn0 + wrow + mi * 16 + crow + r;
+        int k = bk0 + wcol + ni * 16 + ccol;
+        if (oc < d.OC && k < K) dw[(int64_t)oc * K + k] = acc[mi][ni][r];
+      }
+  if (db && blockIdx.y == 0 && tid < 64 && bn0 + tid < d.OC)
+    db[bn0 + tid] = dbs[tid];
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers
+// ---------------------------------------------------------------------------
+
+inline hipStream_t stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+extern bool* g_bf16_flag2;
+
+ConvDims dims_of(const torch::Tensor& x, const torch::Tensor& w, int64_t s) {
+  ConvDims d;
+  d.B = (int)x.size(0); d.IC = (int)x.size(1);
+  d.IH = (int)x.size(2); d.IW = (int)x.size(3);
+  d.OC = (int)w.size(0); d.KH = (int)w.size(2); d.KW = (int)w.size(3);
+  d.S = (int)s;
+  d.OH = (d.IH - d.KH) / d.S + 1;
+  d.OW = (d.IW - d.KW) / d.S + 1;
+  return d;
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, int64_t s,
+                         bool relu) {
+  auto d = dims_of(x, w, s);
+  auto y = torch::empty({d.B, d.OC, d.OH, d.OW}, x.options());
+  const int M = d.B * d.OH * d.OW;
+  dim3 grid((M + TB - 1) / TB, (d.OC + TB - 1) / TB);
+  const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  const bool bf16 = *g_bf16_flag2;
+  auto L = [&](auto b16, auto rl) {
+    hipLaunchKernelGGL((conv_fwd_kernel<decltype(b16)::value,
+                                        decltype(rl)::value>),
+                       grid, dim3(256), 0, stream(), x.data_ptr<float>(),
+                       w.data_ptr<float>(), bp, y.data_ptr<float>(), d);
+  };
+  if (bf16) { if (relu) L(std::true_type{}, std::true_type{});
+              else L(std::true_type{}, std::false_type{}); }
+  else      { if (relu) L(std::false_type{}, std::true_type{});
+              else L(std::false_type{}, std::false_type{}); }
+  return y;
+}
+
+torch::Tensor conv2d_dgrad(torch::Tensor dy,
+                           c10::optional<torch::Tensor> ymask,
+                           torch::Tensor wt, torch::Tensor x_like,
+                           torch::Tensor w, int64_t s) {
+  auto d = dims_of(x_like, w, s);
+  auto dx = torch::empty_like(x_like);
+  const int M = d.B * d.IH * d.IW;
+  dim3 grid((M + TB - 1) / TB, (d.IC + TB - 1) / TB);
+  const bool mask = ymask.has_value();
+  const float* mp = mask ? ymask->data_ptr<float>() : nullptr;
+  const bool bf16 = *g_bf16_flag2;
+  auto L = [&](auto b16, auto mk) {
+    hipLaunchKernelGGL((conv_dgrad_kernel<decltype(b16)::value,
+                                          decltype(mk)::value>),
+                       grid, dim3(256), 0, stream(), dy.data_ptr<float>(),
+                       mp, wt.data_ptr<float>(), dx.data_ptr<float>(), d);
+  };
+  if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
+              else L(std::true_type{}, std::false_type{}); }
+  else      { if (mask) L(std::false_type{}, std::true_type{});
+              else L(std::false_type{}, std::false_type{}); }
+  return dx;
+}
+
+std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
+                                        c10::optional<torch::Tensor> ymask,
+                                        torch::Tensor x, torch::Tensor w,
+                                        int64_t s) {
+  auto d = dims_of(x, w, s);
+  auto dw = torch::empty_like(w);
+  auto db = torch::empty({d.OC}, w.options());
+  const int K = d.IC * d.KH * d.KW;
+  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB);
+  const bool mask = ymask.has_value();
+  const float* mp = mask ? ymask->data_ptr<float>() : nullptr;
+  const bool bf16 = *g_bf16_flag2;
+  auto L = [&](auto b16, auto mk) {
+    hipLaunchKernelGGL((conv_wgrad_kernel<decltype(b16)::value,
+                                          decltype(mk)::value>),
+                       grid, dim3(256), 0, stream(), dy.data_ptr<float>(),
+                       mp, x.data_ptr<float>(), dw.data_ptr<float>(),
+                       db.data_ptr<float>(), d);
+  };
+  if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
+              else L(std::true_type{}, std::false_type{}); }
+  else      { if (mask) L(std::false_type{}, std::true_type{});
+              else L(std::false_type{}, std::false_type{}); }
+  return {dw, db};
+}
+
+}  // namespace convk
+
+namespace convk { bool* g_bf16_flag2 = nullptr; }
+
+void set_conv_bf16_flag(bool* p) { convk::g_bf16_flag2 = p; }
+
+void register_conv(pybind11::module_& m) {
+  m.def("conv2d_fwd", &convk::conv2d_fwd);
+  m.def("conv2d_dgrad", &convk::conv2d_dgrad);
+  m.def("conv2d_wgrad", &convk::conv2d_wgrad);
+}

@@ -109,6 +109,44 @@ def mlp_forward(x, layers, relu_last: bool = True):
 
 
 # ---------------------------------------------------------------------------
+# Conv2d (implicit-GEMM MFMA; valid padding, square stride — the visual
+# trunk's shape family, reference networks/convolutional.py:30-51)
+# ---------------------------------------------------------------------------
+
+class _NativeConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, stride):
+        ext = require_extension()
+        y = ext.conv2d_fwd(x.contiguous(), w.contiguous(),
+                           b.contiguous() if b is not None else None,
+                           stride, False)
+        ctx.save_for_backward(x, w)
+        ctx.stride = stride
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        ext = require_extension()
+        dy = dy.contiguous()
+        dx = None
+        if ctx.needs_input_grad[0]:
+            oc, ic, kh, kw = w.shape
+            wt = w.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw).contiguous()
+            dx = ext.conv2d_dgrad(dy, None, wt, x, w, ctx.stride)
+        dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
+        return dx, dw, (db if ctx.has_bias else None), None
+
+
+def conv2d(x, w, b, stride: int):
+    """Valid-padding conv with square stride (the reference CNN family)."""
+    if use_native(x, w):
+        return _NativeConv2d.apply(x, w, b, int(stride))
+    return F.conv2d(x, w, b, stride=stride)
+
+
+# ---------------------------------------------------------------------------
 # Fused tanh-Gaussian head (sample + squash + log-prob)
 # ---------------------------------------------------------------------------
 

@@ -54,7 +54,7 @@ def test_conv_fwd_parity(ext, B, IC, IH, OC, KH, S):
 def test_conv_bwd_parity(ext, B, IC, IH, OC, KH, S):
     torch.manual_seed(1)
     x = torch.randn(B, IC, IH, IH, requires_grad=True)
-    w = torch.randn(OC, IC, KH, KH, requires_grad=True) / KH
+    w = (torch.randn(OC, IC, KH, KH) / KH).requires_grad_(True)
     b = torch.randn(OC, requires_grad=True)
     y = F.conv2d(x, w, b, stride=S)
     dy = torch.randn_like(y)

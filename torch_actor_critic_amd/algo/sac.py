@@ -43,6 +43,7 @@ from ..optim import FlatAdam
 from ..parallel import comm
 from ..parallel.flat import FlatParams, flatten_module_like
 from ..utils import checkpoint as ckpt
+from ..utils.profiling import Throughput, roctx_range
 
 logger = logging.getLogger(__name__)
 
@@ -346,7 +347,9 @@ class SAC:
         pbar = tqdm.trange(start_epoch, start_epoch + self.epochs, ncols=0,
                            initial=start_epoch, disable=not rank0)
         metrics = {"episode_length": 0.0, "reward": 0.0,
-                   "loss_q": 0.0, "loss_pi": 0.0}
+                   "loss_q": 0.0, "loss_pi": 0.0,
+                   "updates_per_sec": 0.0, "env_steps_per_sec": 0.0}
+        thr = Throughput()
 
         for e in pbar:
             episode_rewards: t.List[float] = []
@@ -365,6 +368,7 @@ class SAC:
                     action = self._select_action(actor, state, device)
 
                 next_state, reward, done, _info = env.step(action)
+                thr.tick_env()
                 ep_len += 1
                 ep_ret += float(reward)
                 done = False if ep_len == self.max_ep_len else done
@@ -393,8 +397,10 @@ class SAC:
                         actor, critic, target_critic, buffer, pi_opt, q_opt,
                         device)
                     if graph is not None:
-                        for _u in range(self.update_every):
-                            graph.step()
+                        with roctx_range("sac_update_burst"):
+                            for _u in range(self.update_every):
+                                graph.step()
+                        thr.tick_update(self.update_every)
                         n_updates += self.update_every
                     else:
                         for _u in range(self.update_every):
@@ -425,6 +431,8 @@ class SAC:
             if all_rews:
                 metrics["reward"] = float(np.mean(all_rews))
                 metrics["episode_length"] = float(np.mean(all_lens))
+            metrics.update(thr.rates())
+            thr.reset()
 
             if rank0 and logging:
                 if (e + 1) % self.save_every == 0:

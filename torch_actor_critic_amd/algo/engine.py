@@ -78,8 +78,7 @@ class FusedSACEngine:
         # actor trunk
         self.a_hidden = [l.out_features for l in actor.layers]
         self.a_act = [torch.zeros(2 * B, h, **f32) for h in self.a_hidden]
-        self.mu = torch.zeros(2 * B, A, **f32)
-        self.ls = torch.zeros(2 * B, A, **f32)
+        self.hl = torch.zeros(2 * B, 2 * A, **f32)   # [mu | log_std]
         self.prob = torch.zeros(2 * B, A, **f32)
         self.logp = torch.zeros(2 * B, **f32)
         self.dmu = torch.zeros(B, A, **f32)
@@ -135,6 +134,16 @@ class FusedSACEngine:
         self._c_offs = offs(q_opt, self._c_tr_src)
         self._a_offs = offs(pi_opt, self._a_tr_src)
 
+        # whole-MLP fused forward feasibility (LDS budget)
+        from ..ops import functional as Fo
+        bf16 = Fo.get_compute_dtype() == "bf16"
+        self._empty = torch.empty(0, **f32)
+        self.use_mlpf = (
+            self.ext.mlp_fwd_fits(OC, list(self.c_w), bf16)
+            and self.ext.mlp_fwd_fits(O, list(self.a_hidden) + [2 * A],
+                                      bf16)
+            and max(self.a_hidden + [2 * A] + self.c_w) <= 256)
+
         # ---- losses / alpha -------------------------------------------
         self.loss_q_acc = torch.zeros(1, **f32)
         self.loss_pi_acc = torch.zeros(1, **f32)
@@ -172,10 +181,25 @@ class FusedSACEngine:
                        xs2 or [], ws2 or [], masks2 or [], K2, x_off,
                        x2_off)
 
-    def _critic_fwd(self, x_src, x_off, weights, acts, lda):
+    def _critic_fwd(self, x_src, x_off, weights, acts, lda,
+                    need_hidden_acts=True):
         """Twin-critic forward; acts[z][i] filled.  Final layer no relu."""
         B = self.B
         nL = len(self.c_w)
+        if self.use_mlpf:
+            e = self._empty
+            act_arg = [
+                [acts[z][i] if (need_hidden_acts or i == nL - 1) else e
+                 for i in range(nL)] for z in range(2)]
+            self.ext.mlp_fwd_fused(
+                x_src, x_off, lda, B, self.OC,
+                [[weights[z][i][0] for i in range(nL)] for z in range(2)],
+                [[e] * nL for _ in range(2)],
+                [[weights[z][i][1] for i in range(nL)] for z in range(2)],
+                [[e] * nL for _ in range(2)],
+                act_arg, list(self.c_w), [1 << 30] * nL,
+                (1 << (nL - 1)) - 1)
+            return
         x = [x_src, x_src]
         k = self.OC
         src_off = x_off
@@ -203,23 +227,39 @@ class FusedSACEngine:
                         self.XC, self.XC2, self.rew, self.done, B)
 
         # actor forward over stacked 2B rows of XC[:, :O]
-        x, k, lda, off = self.XC, O, OC, 0
-        for i, (w, b) in enumerate(self.aw):
-            self._mg([x], [w], [b], [self.a_act[i]], [None], 2 * B,
-                     self.a_hidden[i], k, lda, self.a_hidden[i], True,
-                     x_off=off)
-            x, k, lda, off = self.a_act[i], self.a_hidden[i], \
-                self.a_hidden[i], 0
         (wm, bm), (wl, bl) = self.head_w
-        self._mg([x, x], [wm, wl], [bm, bl], [self.mu, self.ls],
-                 [None, None], 2 * B, A, k, lda, A, False)
+        if self.use_mlpf:
+            e = self._empty
+            nT = len(self.aw)
+            ws = [[w for (w, _) in self.aw] + [wm]]
+            whis = [[e] * nT + [wl]]
+            bs = [[b for (_, b) in self.aw] + [bm]]
+            bhis = [[e] * nT + [bl]]
+            acts = [list(self.a_act) + [self.hl]]
+            widths = list(self.a_hidden) + [2 * A]
+            splits = [1 << 30] * nT + [A]
+            relu_mask = (1 << nT) - 1
+            ext.mlp_fwd_fused(self.XC, 0, OC, 2 * B, O, ws, whis, bs,
+                              bhis, acts, widths, splits, relu_mask)
+        else:
+            x, k, lda, off = self.XC, O, OC, 0
+            for i, (w, b) in enumerate(self.aw):
+                self._mg([x], [w], [b], [self.a_act[i]], [None], 2 * B,
+                         self.a_hidden[i], k, lda, self.a_hidden[i], True,
+                         x_off=off)
+                x, k, lda, off = self.a_act[i], self.a_hidden[i], \
+                    self.a_hidden[i], 0
+            self._mg([x, x], [wm, wl], [bm, bl],
+                     [self.hl, self.hl[:, A:]],
+                     [None, None], 2 * B, A, k, lda, 2 * A, False)
         # pi rows :B -> XC2[:, O:], a2 rows B: -> XC[B:, O:]
-        ext.tg_fwd2(self.mu, self.ls, self.XC2, O, self.XC, B * OC + O,
+        ext.tg_fwd2(self.hl, self.XC2, O, self.XC, B * OC + O,
                     B, self.logp, self.prob, self.ctr, self.seed,
                     self.act_limit, self.lo, self.hi)
 
         # target critic on (ns, a2) = XC rows B:
-        self._critic_fwd(self.XC, B * OC, self.tw, self.t_act, OC)
+        self._critic_fwd(self.XC, B * OC, self.tw, self.t_act, OC,
+                         need_hidden_acts=False)
         # critic on (s, a) = XC rows :B
         self._critic_fwd(self.XC, 0, self.cw, self.c_act, OC)
 
@@ -293,7 +333,7 @@ class FusedSACEngine:
 
         # actor backward
         ext.tg_bwd2(self.dxc, O, self.alpha_dev, self.alpha_host,
-                    self.mu, self.ls, self.prob, self.dmu, self.dls, B,
+                    self.hl, self.prob, self.dmu, self.dls, B,
                     self.act_limit, self.lo, self.hi)
         h_last = self.a_hidden[-1]
         a_last = self.a_act[-1]

@@ -636,8 +636,7 @@ void gather2_kernel(const float* __restrict__ state,
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(64)
-void tg_fwd2_kernel(const float* __restrict__ mu,
-                    const float* __restrict__ ls_raw,
+void tg_fwd2_kernel(const float* __restrict__ hl,  // [R, 2A]: mu | log_std
                     float* __restrict__ out0, int ld0,
                     float* __restrict__ out1, int ld1, int B0,
                     float* __restrict__ logp, float* __restrict__ prob_out,
@@ -655,8 +654,8 @@ void tg_fwd2_kernel(const float* __restrict__ mu,
     float u1 = (r.y + 1.f) * 2.3283064365386963e-10f;
     float eps = sqrtf(-2.f * logf(u0)) * __cosf(6.283185307179586f * u1);
 
-    float m = mu[i];
-    float ls = fminf(fmaxf(ls_raw[i], lo), hi);
+    float m = hl[(int64_t)b * 2 * A + a];
+    float ls = fminf(fmaxf(hl[(int64_t)b * 2 * A + A + a], lo), hi);
     float std = expf(ls);
     float prob = m + std * eps;
     float pi = tanhf(prob) * act_limit;
@@ -695,8 +694,7 @@ void tg_eps_kernel(float* __restrict__ eps_out, uint64_t ctr_val,
 __global__ __launch_bounds__(64)
 void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
                     const float* __restrict__ alpha_dev, float alpha_host,
-                    const float* __restrict__ mu,
-                    const float* __restrict__ ls_raw,
+                    const float* __restrict__ hl,   // [R, 2A]: mu | log_std
                     const float* __restrict__ prob,
                     float* __restrict__ dmu, float* __restrict__ dls,
                     int B, int A, float act_limit, float lo, float hi) {
@@ -705,12 +703,12 @@ void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
   if (a >= A) return;
   const int64_t i = (int64_t)b * A + a;
   const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
-  float raw = ls_raw[i];
+  float raw = hl[(int64_t)b * 2 * A + A + a];
   float ls = fminf(fmaxf(raw, lo), hi);
   float std = expf(ls);
   float p = prob[i];
   float t = tanhf(p);
-  float se = p - mu[i];              // std * eps
+  float se = p - hl[(int64_t)b * 2 * A + a];   // std * eps
   float dpi = dxc[(int64_t)b * ld_dxc + col0 + a];
   float dl = alpha / B;
   float dp = dpi * act_limit * (1.f - t * t);
@@ -993,14 +991,14 @@ void gather2(torch::Tensor state, torch::Tensor act, torch::Tensor rew,
                      (int)B, obs_dim, act_dim);
 }
 
-void tg_fwd2(torch::Tensor mu, torch::Tensor ls, torch::Tensor out0,
+void tg_fwd2(torch::Tensor hl, torch::Tensor out0,
              int64_t col0, torch::Tensor out1, int64_t col1, int64_t B0,
              torch::Tensor logp, torch::Tensor prob, torch::Tensor ctr,
              int64_t seed, double act_limit, double lo, double hi) {
-  const int R = (int)mu.size(0), A = (int)mu.size(1);
+  const int R = (int)hl.size(0), A = (int)hl.size(1) / 2;
   TORCH_CHECK(A <= 64);
   hipLaunchKernelGGL(tg_fwd2_kernel, dim3(R), dim3(64), 0, stream(),
-                     mu.data_ptr<float>(), ls.data_ptr<float>(),
+                     hl.data_ptr<float>(),
                      out0.data_ptr<float>() + col0, (int)out0.size(1),
                      out1.data_ptr<float>() + col1, (int)out1.size(1),
                      (int)B0, logp.data_ptr<float>(), prob.data_ptr<float>(),
@@ -1010,14 +1008,14 @@ void tg_fwd2(torch::Tensor mu, torch::Tensor ls, torch::Tensor out0,
 
 void tg_bwd2(torch::Tensor dxc, int64_t col0,
              c10::optional<torch::Tensor> alpha_dev, double alpha_host,
-             torch::Tensor mu, torch::Tensor ls, torch::Tensor prob,
+             torch::Tensor hl, torch::Tensor prob,
              torch::Tensor dmu, torch::Tensor dls, int64_t B,
              double act_limit, double lo, double hi) {
   const int A = (int)dmu.size(1);
   hipLaunchKernelGGL(tg_bwd2_kernel, dim3((int)B), dim3(64), 0, stream(),
                      dxc.data_ptr<float>(), (int)dxc.size(1), (int)col0,
                      fptr(alpha_dev), (float)alpha_host,
-                     mu.data_ptr<float>(), ls.data_ptr<float>(),
+                     hl.data_ptr<float>(),
                      prob.data_ptr<float>(), dmu.data_ptr<float>(),
                      dls.data_ptr<float>(), (int)B, A, (float)act_limit,
                      (float)lo, (float)hi);

@@ -1031,12 +1031,16 @@ namespace fused { bool* g_bf16_flag = nullptr; }
 void register_fused(pybind11::module_& m);
 void register_conv(pybind11::module_& m);
 void set_conv_bf16_flag(bool* p);
+void register_mlpf(pybind11::module_& m);
+void set_mlpf_bf16_flag(bool* p);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   fused::g_bf16_flag = &g_bf16;
   set_conv_bf16_flag(&g_bf16);
+  set_mlpf_bf16_flag(&g_bf16);
   register_fused(m);
   register_conv(m);
+  register_mlpf(m);
   m.def("set_compute_bf16", &set_compute_bf16,
         "Switch GEMM kernels to bf16 MFMA inputs (fp32 accumulate)");
   m.def("get_compute_bf16", &get_compute_bf16);

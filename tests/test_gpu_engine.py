@@ -127,7 +127,7 @@ def test_engine_full_update_parity():
     torch.cuda.synchronize()
 
     # the engine bumped its counter once -> noise drawn at ctr=1
-    eps = ext.tg_eps(1, 0, 2 * B, A, eng.mu).cpu()
+    eps = ext.tg_eps(1, 0, 2 * B, A, eng.prob).cpu()
     ref = _eager_reference(actor_cpu, critic_cpu, target_cpu,
                            s, a, r, ns, d, eps)
 

@@ -100,21 +100,44 @@ void mlp_fwd_kernel(Args a) {
     if (wc0 < H) {
       f32x4 acc[4][4] = {};
       const int kpad = ((K + BK - 1) / BK) * BK;
-      for (int k0 = 0; k0 < kpad; k0 += BK) {
-        // stage this wave's 64 weight rows x BK cols (lane = row)
-        {
-          const int row = wc0 + lane;
-          int rr;
-          const float* wsrc = row < H
-              ? wrow_ptr<BF16>(a, z, L, row, &rr) : nullptr;
-#pragma unroll 4
-          for (int c = 0; c < BK; ++c) {
-            float v = 0.f;
-            if (wsrc && k0 + c < K)
-              v = wsrc[(int64_t)rr * K + k0 + c];
-            WS[lane * a.ldsw + c] = (T)v;
+      const int row = wc0 + lane;
+      int rr = 0;
+      const float* wsrc = row < H
+          ? wrow_ptr<BF16>(a, z, L, row, &rr) : nullptr;
+      const int Kc = K;
+      float vw[BK];
+      auto load_w = [&](int k0) {
+        if (wsrc && ((Kc & 3) == 0) && (k0 + BK <= Kc)) {
+          const float4* s = (const float4*)(wsrc + (int64_t)rr * Kc + k0);
+#pragma unroll
+          for (int q = 0; q < BK / 4; ++q) {
+            float4 f = s[q];
+            vw[q*4+0]=f.x; vw[q*4+1]=f.y; vw[q*4+2]=f.z; vw[q*4+3]=f.w;
           }
+        } else {
+#pragma unroll
+          for (int c = 0; c < BK; ++c)
+            vw[c] = (wsrc && k0 + c < Kc)
+                        ? wsrc[(int64_t)rr * Kc + k0 + c] : 0.f;
         }
+      };
+      load_w(0);
+      for (int k0 = 0; k0 < kpad; k0 += BK) {
+        // write the prefetched W chunk to this wave's LDS strip
+        if constexpr (BF16) {
+          union { __bf16 h[BK]; uint4 u[BK / 8]; } pk;
+#pragma unroll
+          for (int c = 0; c < BK; ++c) pk.h[c] = (__bf16)vw[c];
+          uint4* dst = (uint4*)&WS[lane * a.ldsw];
+#pragma unroll
+          for (int q = 0; q < BK / 8; ++q) dst[q] = pk.u[q];
+        } else {
+          float* dst = (float*)&WS[lane * a.ldsw];
+#pragma unroll
+          for (int c = 0; c < BK; ++c) dst[c] = vw[c];
+        }
+        // prefetch the next chunk while the MFMAs run (T14)
+        if (k0 + BK < kpad) load_w(k0 + BK);
         // wave-private strip: no barrier needed; MFMA over the chunk
         if constexpr (BF16) {
           const int arow = lane & 15;

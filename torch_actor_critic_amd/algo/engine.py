@@ -138,8 +138,16 @@ class FusedSACEngine:
         from ..ops import functional as Fo
         bf16 = Fo.get_compute_dtype() == "bf16"
         self._empty = torch.empty(0, **f32)
+        # Whole-MLP fusion only pays when there are enough 64-row blocks
+        # to fill the 256 CUs (measured: at B=64 the layered pipelined
+        # GEMMs win — 8 N-parallel blocks vs 2 fused blocks); threshold
+        # overridable with TAC_AMD_MLPF={0,1}.
+        import os
+        mlpf_env = os.environ.get("TAC_AMD_MLPF")
+        big_enough = B >= 1024 if mlpf_env is None else mlpf_env == "1"
         self.use_mlpf = (
-            self.ext.mlp_fwd_fits(OC, list(self.c_w), bf16)
+            big_enough
+            and self.ext.mlp_fwd_fits(OC, list(self.c_w), bf16)
             and self.ext.mlp_fwd_fits(O, list(self.a_hidden) + [2 * A],
                                       bf16)
             and max(self.a_hidden + [2 * A] + self.c_w) <= 256)

@@ -138,15 +138,14 @@ class FusedSACEngine:
         from ..ops import functional as Fo
         bf16 = Fo.get_compute_dtype() == "bf16"
         self._empty = torch.empty(0, **f32)
-        # Whole-MLP fusion only pays when there are enough 64-row blocks
-        # to fill the 256 CUs (measured: at B=64 the layered pipelined
-        # GEMMs win — 8 N-parallel blocks vs 2 fused blocks); threshold
-        # overridable with TAC_AMD_MLPF={0,1}.
+        # Whole-MLP fusion measured SLOWER than the layered pipelined
+        # GEMMs at both batch 64 (2388 vs 4644 upd/s) and batch 4096
+        # (683 vs 777): the layered path's N/M-parallel grid beats the
+        # fused kernel's per-block layer serialization.  Kept available
+        # for study via TAC_AMD_MLPF=1; default off.
         import os
-        mlpf_env = os.environ.get("TAC_AMD_MLPF")
-        big_enough = B >= 1024 if mlpf_env is None else mlpf_env == "1"
         self.use_mlpf = (
-            big_enough
+            os.environ.get("TAC_AMD_MLPF") == "1"
             and self.ext.mlp_fwd_fits(OC, list(self.c_w), bf16)
             and self.ext.mlp_fwd_fits(O, list(self.a_hidden) + [2 * A],
                                       bf16)

@@ -1,0 +1,67 @@
+"""CLI integration tests: train -> resume -> evaluate through the real
+main.py / run_agent.py entry points (reference CLI surface,
+main.py:113-125, run_agent.py:51-59)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(args, cwd, timeout=240):
+    return subprocess.run([sys.executable] + args, cwd=cwd, timeout=timeout,
+                          capture_output=True, text=True,
+                          env={**os.environ, "PYTHONPATH": REPO})
+
+
+@pytest.mark.timeout(600)
+def test_train_resume_eval_roundtrip(tmp_path):
+    # 1) short training run (10 epochs so save_every=10 checkpoints)
+    r = _run([os.path.join(REPO, "main.py"), "--environment", "Pendulum-v1",
+              "--epochs", "10", "--steps-per-epoch", "120",
+              "--batch-size", "32", "--buffer-size", "5000",
+              "--device", "cpu"], cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+
+    runs = os.listdir(tmp_path / "mlruns" / "0")
+    assert len(runs) == 1
+    run_id = runs[0]
+    art = tmp_path / "mlruns" / "0" / run_id / "artifacts"
+    assert (art / "actor" / "data" / "model.pth").exists()
+    assert (art / "auxiliaries" / "state_dict.pth").exists()
+    # reward metric logged
+    assert (tmp_path / "mlruns" / "0" / run_id / "metrics" /
+            "reward").exists()
+
+    # 2) resume from the run
+    r = _run([os.path.join(REPO, "main.py"), "--run", run_id,
+              "--epochs", "1", "--steps-per-epoch", "200",
+              "--buffer-size", "5000", "--device", "cpu"], cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+
+    # 3) evaluate with run_agent.py
+    r = _run([os.path.join(REPO, "run_agent.py"), "--run", run_id,
+              "--episodes", "2", "--headless", "--device", "cpu"],
+             cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+@pytest.mark.timeout(300)
+def test_bench_contract_json(tmp_path):
+    r = _run([os.path.join(REPO, "bench.py"), "--steps", "30",
+              "--warmup", "5"], cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in d, key
+    assert d["metric"] == "sac_updates_per_sec"
+    assert d["value"] > 0
+    assert d["data"] == "synthetic"
+    assert d["config"]["parallelism"] == "dp1"

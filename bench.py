@@ -71,14 +71,24 @@ def main():
 
     env = envs.make(args.env)
     env.seed(1000 + rank)
+    visual = args.env in ("DeepMindWallRunner-v0", "VisualCheetahRun-v0")
     obs_dim = env.observation_space.shape[0]
     act_dim = env.action_space.shape[0]
     act_limit = float(env.action_space.high.reshape(-1)[0])
 
     torch.manual_seed(0)  # identical init on all ranks
-    actor = Actor(obs_dim, act_dim, list(args.hidden),
-                  act_limit=act_limit).to(device)
-    critic = DoubleCritic(obs_dim, act_dim, list(args.hidden)).to(device)
+    if visual:
+        from torch_actor_critic_amd.models.visual import (
+            VisualActor, VisualDoubleCritic)
+        vis_dim = tuple(env.vis_dim)
+        actor = VisualActor(obs_dim, act_dim, vis_dim, list(args.hidden),
+                            act_limit).to(device)
+        critic = VisualDoubleCritic(obs_dim, act_dim, vis_dim,
+                                    list(args.hidden)).to(device)
+    else:
+        actor = Actor(obs_dim, act_dim, list(args.hidden),
+                      act_limit=act_limit).to(device)
+        critic = DoubleCritic(obs_dim, act_dim, list(args.hidden)).to(device)
     target_critic = deepcopy(critic)
     _freeze(target_critic, True)
 
@@ -96,26 +106,48 @@ def main():
               update_every=args.update_window, save_every=10**9)
 
     # ---- prefill the HBM replay ring with synthetic transitions -------
-    buf_size = args.buffer_size if on_gpu else min(args.buffer_size, 100_000)
-    buffer = ReplayBuffer(buf_size, obs_dim, act_dim, device=device,
-                          seed=rank)
     rng = np.random.default_rng(rank)
-    prefill = max(10_000, 4 * args.batch_size)
-    chunk = 10_000
-    for i0 in range(0, prefill, chunk):
-        n = min(chunk, prefill - i0)
-        buffer.store_batch(
-            rng.standard_normal((n, obs_dim)).astype(np.float32),
-            rng.standard_normal((n, act_dim)).astype(np.float32),
-            rng.standard_normal(n).astype(np.float32),
-            rng.standard_normal((n, obs_dim)).astype(np.float32),
-            np.zeros(n, dtype=np.float32))
+    if visual:
+        from torch_actor_critic_amd.buffer.visual import VisualReplayBuffer
+        # 100k visual transitions (u8 frames) fit comfortably in HBM
+        buf_size = min(args.buffer_size, 100_000) if on_gpu else 2_000
+        buffer = VisualReplayBuffer(buf_size, act_dim, device=device,
+                                    seed=rank)
+        buffer.store(env.reset(), np.zeros(act_dim), 0.0, env.reset(), 0.0)
+        prefill = min(buf_size, max(4096, 4 * args.batch_size))
+        # fill the dense ring tensors directly (synthetic data)
+        buffer.features[:prefill].normal_()
+        buffer.next_features[:prefill].normal_()
+        buffer.frames[:prefill].random_(0, 255)
+        buffer.next_frames[:prefill].random_(0, 255)
+        buffer.actions[:prefill].uniform_(-1, 1)
+        buffer.rewards[:prefill].normal_()
+        buffer.ptr = prefill % buf_size
+        buffer.size = prefill
+        buffer._size_dev.fill_(prefill)
+    else:
+        buf_size = args.buffer_size if on_gpu else min(args.buffer_size,
+                                                       100_000)
+        buffer = ReplayBuffer(buf_size, obs_dim, act_dim, device=device,
+                              seed=rank)
+        prefill = max(10_000, 4 * args.batch_size)
+        chunk = 10_000
+        for i0 in range(0, prefill, chunk):
+            n = min(chunk, prefill - i0)
+            buffer.store_batch(
+                rng.standard_normal((n, obs_dim)).astype(np.float32),
+                rng.standard_normal((n, act_dim)).astype(np.float32),
+                rng.standard_normal(n).astype(np.float32),
+                rng.standard_normal((n, obs_dim)).astype(np.float32),
+                np.zeros(n, dtype=np.float32))
 
     # ---- build the update path (hipGraph on GPU) ----------------------
     graph = None
     engine_kind = "eager"
     if on_gpu and not args.no_graph:
         try:
+            if visual:
+                raise RuntimeError("visual models use the autograd graph")
             from torch_actor_critic_amd.algo.engine import FusedSACEngine
             graph = FusedSACEngine(sac, actor, critic, target_critic,
                                    buffer, pi_opt, q_opt, target_flat,
@@ -138,7 +170,7 @@ def main():
     state = env.reset()
 
     act_graph = None
-    if on_gpu:
+    if on_gpu and not visual:
         from torch_actor_critic_amd.algo.act import ActGraph, WindowedStore
         act_graph = ActGraph(actor, obs_dim, act_dim, device)
         wstore = WindowedStore(buffer, args.update_window)
@@ -159,10 +191,16 @@ def main():
                     state = env.reset() if done else nstate
                 wstore.flush()
             else:
+                from torch_actor_critic_amd.envs.visual import (
+                    MultiObservation)
                 with torch.no_grad():
                     for _ in range(n_steps):
-                        s = torch.as_tensor(state, dtype=torch.float32,
-                                            device=device)
+                        if isinstance(state, MultiObservation):
+                            s = MultiObservation(state.features.to(device),
+                                                 state.frame.to(device))
+                        else:
+                            s = torch.as_tensor(state, dtype=torch.float32,
+                                                device=device)
                         a, _ = actor(s, deterministic=False,
                                      with_logprob=False)
                         a_np = a.detach().cpu().numpy()
@@ -231,7 +269,7 @@ def main():
             "dtype": dtype,
             "data": "synthetic",
             "config": {
-                "model": f"SAC-MLP{args.hidden}-{args.env}",
+                "model": (f"SAC-CNN+MLP{args.hidden}-{args.env}" if visual else f"SAC-MLP{args.hidden}-{args.env}"),
                 "global_batch": args.batch_size * world,
                 "seq_len": 1,
                 "parallelism": f"dp{world}",

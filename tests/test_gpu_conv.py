@@ -160,3 +160,58 @@ def test_visual_sac_update_on_gpu():
     assert np.isfinite(float(loss_pi.item()))
     assert torch.isfinite(q_opt.fp.flat_grad).all()
     assert torch.isfinite(pi_opt.fp.flat_grad).all()
+
+
+def test_visual_graphed_update_on_gpu():
+    """The autograd hipGraph path captures the full visual SAC update
+    (conv fwd/bwd kernels inside the graph) and replays it."""
+    from copy import deepcopy
+    from torch_actor_critic_amd.algo.graph import GraphedSACUpdate
+    from torch_actor_critic_amd.algo.sac import SAC, _freeze
+    from torch_actor_critic_amd.buffer.visual import VisualReplayBuffer
+    from torch_actor_critic_amd.envs.visual import MultiObservation
+    from torch_actor_critic_amd.models.visual import (VisualActor,
+                                                      VisualDoubleCritic)
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.parallel.flat import flatten_module_like
+
+    torch.manual_seed(5)
+    device = torch.device(DEV)
+    actor = VisualActor(16, 4, (3, 64, 64), [32, 32],
+                        act_limit=1.0).to(device)
+    critic = VisualDoubleCritic(16, 4, (3, 64, 64), [32, 32]).to(device)
+    target = deepcopy(critic)
+    _freeze(target, True)
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+    target_flat = flatten_module_like(target)
+
+    buf = VisualReplayBuffer(200, act_dim=4, device=device)
+    mo = MultiObservation(torch.randn(16, device=device),
+                          torch.randn(3, 64, 64, device=device))
+    buf.store(mo, np.zeros(4), 0.0, mo, 0.0)
+    n = 100
+    buf.features[:n].normal_()
+    buf.next_features[:n].normal_()
+    buf.frames[:n].random_(0, 255)
+    buf.next_frames[:n].random_(0, 255)
+    buf.actions[:n].uniform_(-1, 1)
+    buf.rewards[:n].normal_()
+    buf.size = n
+    buf.ptr = n % buf.max_size
+    buf._size_dev.fill_(n)
+
+    sac = SAC(alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=8, start_steps=0, steps_per_epoch=1,
+              max_ep_len=10, update_after=0, update_every=1, save_every=10)
+    g = GraphedSACUpdate(sac, actor, critic, target, buf, pi_opt, q_opt,
+                         target_flat, 8, device)
+    p0 = pi_opt.fp.flat.clone()
+    t0 = target_flat.clone()
+    for _ in range(5):
+        g.step()
+    torch.cuda.synchronize()
+    assert not torch.allclose(p0, pi_opt.fp.flat)
+    assert not torch.allclose(t0, target_flat)
+    lq, lp = g.read_and_reset_losses(5)
+    assert np.isfinite(lq) and np.isfinite(lp)
+    assert torch.isfinite(pi_opt.fp.flat).all()

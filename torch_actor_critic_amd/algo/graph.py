@@ -50,16 +50,9 @@ class GraphedSACUpdate:
         self.device = device
         self.world = comm.num_procs()
 
-        obs_dim = buffer.obs_dim
-        act_dim = buffer.act_dim
         opts = dict(device=device, dtype=torch.float32)
-        self.batch = Batch(
-            states=torch.zeros(batch_size, obs_dim, **opts),
-            actions=torch.zeros(batch_size, act_dim, **opts),
-            rewards=torch.zeros(batch_size, **opts),
-            next_states=torch.zeros(batch_size, obs_dim, **opts),
-            done=torch.zeros(batch_size, **opts),
-        )
+        # works for both flat (Batch) and visual (VisualBatch) buffers
+        self.batch = buffer.make_static_batch(batch_size)
         # static loss accumulators (read at epoch boundaries only)
         self.loss_q_acc = torch.zeros((), **opts)
         self.loss_pi_acc = torch.zeros((), **opts)

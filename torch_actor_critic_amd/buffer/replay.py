@@ -128,6 +128,15 @@ class ReplayBuffer:
         idx = torch.as_tensor(idx, dtype=torch.long, device=self.device)
         return self.sample_at(idx)
 
+    def make_static_batch(self, batch_size: int) -> Batch:
+        """Preallocated batch tensors for hipGraph-captured sampling."""
+        f32 = dict(dtype=torch.float32, device=self.device)
+        return Batch(torch.zeros(batch_size, self.obs_dim, **f32),
+                     torch.zeros(batch_size, self.act_dim, **f32),
+                     torch.zeros(batch_size, **f32),
+                     torch.zeros(batch_size, self.obs_dim, **f32),
+                     torch.zeros(batch_size, **f32))
+
     def sample_into(self, out: Batch) -> None:
         """Graph-capturable sampling into preallocated batch tensors."""
         ext = self._native_ext()

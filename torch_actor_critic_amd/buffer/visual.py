@@ -45,6 +45,8 @@ class VisualReplayBuffer:
                                    dtype=torch.float32, device=dev)
         self.rewards = torch.zeros(self.max_size, dtype=torch.float32, device=dev)
         self.done = torch.zeros(self.max_size, dtype=torch.float32, device=dev)
+        # device-side valid-size mirror (graph-captured sampling)
+        self._size_dev = torch.zeros(1, dtype=torch.int64, device=dev)
 
     def _alloc(self, obs: MultiObservation):
         feat_dim = int(obs.features.numel())
@@ -85,6 +87,37 @@ class VisualReplayBuffer:
         self.done[i] = float(done)
         self.ptr = (self.ptr + 1) % self.max_size
         self.size = min(self.size + 1, self.max_size)
+        self._size_dev.fill_(self.size)
+
+    def make_static_batch(self, batch_size: int) -> VisualBatch:
+        """Preallocated batch tensors for hipGraph-captured sampling."""
+        assert self._alloc_done, "store at least one transition first"
+        dev = self.device
+        f32 = dict(dtype=torch.float32, device=dev)
+
+        def mo():
+            return MultiObservation(
+                torch.zeros(batch_size, self.feat_dim, **f32),
+                torch.zeros(batch_size, *self.vis_dim, **f32))
+
+        return VisualBatch(mo(), torch.zeros(batch_size, self.act_dim, **f32),
+                           torch.zeros(batch_size, **f32), mo(),
+                           torch.zeros(batch_size, **f32))
+
+    def sample_into(self, out: VisualBatch) -> None:
+        """Graph-capturable sampling: index draw and gathers are pure
+        tensor ops (torch RNG is hipGraph-aware)."""
+        B = out.actions.shape[0]
+        u = torch.rand(B, device=self.device)
+        size = self._size_dev.clamp(min=1).to(torch.float32)
+        idx = (u * size).long().clamp_(max=self.max_size - 1)
+        out.states.features.copy_(self.features[idx])
+        out.states.frame.copy_(self._dec_frames(self.frames[idx]))
+        out.next_states.features.copy_(self.next_features[idx])
+        out.next_states.frame.copy_(self._dec_frames(self.next_frames[idx]))
+        out.actions.copy_(self.actions[idx])
+        out.rewards.copy_(self.rewards[idx])
+        out.done.copy_(self.done[idx])
 
     def sample(self, batch_size: int) -> VisualBatch:
         idx_np = self._rng.choice(self.size, size=batch_size, replace=False)

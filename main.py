@@ -122,6 +122,10 @@ def parse_arguments() -> Namespace:
     parser.add_argument("--reference-pi-loss", action="store_true",
                         help="Reproduce the reference's next_state policy "
                              "loss quirk (SURVEY.md Q2)")
+    parser.add_argument("--normalize-states", action="store_true",
+                        help="Welford online state normalization (wires in "
+                             "the reference's dead sac/utils.py normalizer "
+                             "— SURVEY.md Q9)")
     parser.set_defaults(logging=True, render=False)
     return parser.parse_args()
 
@@ -180,6 +184,10 @@ def main():
 
     sac = SAC(**params, learn_alpha=args.learn_alpha,
               reference_pi_loss=args.reference_pi_loss)
+    if args.normalize_states:
+        from torch_actor_critic_amd.utils.normalizer import (
+            WelfordVarianceEstimate)
+        sac.normalizer = WelfordVarianceEstimate()
     env = envs.make(args.environment)
 
     try:

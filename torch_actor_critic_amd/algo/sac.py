@@ -141,6 +141,11 @@ class SAC:
         self._actor_fp: t.Optional[FlatParams] = None
         self._graph = None
         self._graph_failed = False
+        # optional state normalizer (reference ships one as dead code,
+        # SURVEY.md Q9; wire-in via main.py --normalize-states).  Note:
+        # normalization happens on the acting path only (stored states
+        # are normalized), so the update path is unaffected.
+        self.normalizer = None
 
     # -- single-module updates (reference method surface) ---------------
 
@@ -259,11 +264,22 @@ class SAC:
         critic_cpu = deepcopy(critic).cpu()
         ckpt.log_model(actor_cpu, "actor")
         ckpt.log_model(critic_cpu, "critic")
-        ckpt.log_state_dict({
+        aux = {
             "pi_opt": pi_opt.state_dict(),
             "q_opt": q_opt.state_dict(),
             "epoch": epoch,
-        }, "auxiliaries")
+        }
+        # learned entropy temperature (extension; absent for fixed alpha,
+        # keeping the reference auxiliaries layout otherwise identical)
+        if self.learn_alpha:
+            la = None
+            if self._graph is not None and hasattr(self._graph, "log_alpha"):
+                la = self._graph.log_alpha
+            elif self._log_alpha is not None:
+                la = self._log_alpha
+            if la is not None:
+                aux["log_alpha"] = float(la.detach().cpu().reshape(-1)[0])
+        ckpt.log_state_dict(aux, "auxiliaries")
         del actor_cpu, critic_cpu
         _ = dev
 
@@ -360,6 +376,11 @@ class SAC:
             n_updates = 0
 
             for _ in range(self.steps_per_epoch):
+                if (self.normalizer is not None
+                        and not isinstance(state, MultiObservation)):
+                    st = torch.as_tensor(np.asarray(state, dtype=np.float32))
+                    self.normalizer.update(st)
+                    state = self.normalizer.normalize_state(st).numpy()
                 if step < self.start_steps:
                     action = env.action_space.sample()
                 elif act_graph is not None:

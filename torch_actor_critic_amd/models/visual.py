@@ -42,6 +42,14 @@ def calculate_size(image_shape, filters, kernel_sizes, strides) -> int:
     return int(c * h * w)
 
 
+class Linear(nn.Linear):
+    """nn.Linear whose forward dispatches to the hand-written gfx950
+    MFMA GEMM on GPU; state-dict identical to nn.Linear."""
+
+    def forward(self, x):
+        return Fo.linear_relu(x, self.weight, self.bias, relu=False)
+
+
 class Conv2d(nn.Conv2d):
     """nn.Conv2d whose forward dispatches to the hand-written gfx950
     implicit-GEMM MFMA kernels on GPU (ops/csrc/conv.hip); state-dict
@@ -66,8 +74,8 @@ def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
         model.add_module(f"relu_{i}", activation())
     flat = calculate_size(input_shape, filters, kernel_sizes, strides)
     model.add_module("flatten", nn.Flatten())
-    model.add_module("linear", nn.Linear(flat, dense_size))
-    model.add_module("final", nn.Linear(dense_size, 1))
+    model.add_module("linear", Linear(flat, dense_size))
+    model.add_module("final", Linear(dense_size, 1))
     return model
 
 

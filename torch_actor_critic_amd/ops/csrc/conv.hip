@@ -95,6 +95,32 @@ DEVINL float im2col_at(const float* x, const ConvDims& d, int m, int k) {
   return x[(((int64_t)b * d.IC + ic) * d.IH + iy) * d.IW + ix];
 }
 
+// incremental decompositions (divisions hoisted out of element loops)
+struct MDec { int b, oy, ox; };
+DEVINL MDec mdec(int m, const ConvDims& d) {
+  MDec r;
+  r.ox = m % d.OW;
+  int t = m / d.OW;
+  r.oy = t % d.OH;
+  r.b = t / d.OH;
+  return r;
+}
+DEVINL void minc(MDec& r, const ConvDims& d) {
+  if (++r.ox == d.OW) { r.ox = 0; if (++r.oy == d.OH) { r.oy = 0; ++r.b; } }
+}
+struct KDec { int ic, ky, kx; };
+DEVINL KDec kdec(int k, const ConvDims& d) {
+  KDec r;
+  r.kx = k % d.KW;
+  int t = k / d.KW;
+  r.ky = t % d.KH;
+  r.ic = t / d.KH;
+  return r;
+}
+DEVINL void kinc(KDec& r, const ConvDims& d) {
+  if (++r.kx == d.KW) { r.kx = 0; if (++r.ky == d.KH) { r.ky = 0; ++r.ic; } }
+}
+
 // ---------------------------------------------------------------------------
 // conv fwd: grid (ceil(M/64), ceil(OC/64)); Y NCHW scatter epilogue
 // ---------------------------------------------------------------------------
@@ -123,16 +149,24 @@ void conv_fwd_kernel(const float* __restrict__ x, const float* __restrict__ w,
 
   const int row = tid & 63;
   const int c00 = (tid >> 6) * EL;
+  const int m_my = bm0 + row;
+  const MDec md = mdec(m_my < M ? m_my : 0, d);
   for (int k0 = 0; k0 < K; k0 += BK) {
+    KDec kd = kdec((k0 + c00) < K ? (k0 + c00) : 0, d);
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
       int c = c00 + e;
-      int m = bm0 + row, k = k0 + c;
-      float va = (m < M && k < K) ? im2col_at(x, d, m, k) : 0.f;
+      int k = k0 + c;
+      float va = 0.f;
+      if (m_my < M && k < K) {
+        int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
+        va = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
+      }
       lds_put<BF16>(xs, row, c, va);
       int n = bn0 + row;
       float vb = (n < d.OC && k < K) ? w[(int64_t)n * K + k] : 0.f;
       lds_put<BF16>(ws, row, c, vb);
+      kinc(kd, d);
     }
     __syncthreads();
     mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
@@ -188,22 +222,33 @@ void conv_dgrad_kernel(const float* __restrict__ dy,
 
   const int row = tid & 63;
   const int c00 = (tid >> 6) * EL;
+  const int m_my = bm0 + row;
+  int ix0, iy0, b0;
+  {
+    int m = m_my < M ? m_my : 0;
+    ix0 = m % d.IW;
+    int t1 = m / d.IW;
+    iy0 = t1 % d.IH;
+    b0 = t1 / d.IH;
+  }
   for (int k0 = 0; k0 < K; k0 += BK) {
+    // k = (oc, ky, kx) walked incrementally (KW/KH in place of d's
+    // meanings: here k indexes (oc, ky, kx))
+    int kx = (k0 + c00) % d.KW;
+    int t2 = (k0 + c00) / d.KW;
+    int ky = t2 % d.KH;
+    int oc = t2 / d.KH;
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
       int c = c00 + e;
-      int m = bm0 + row, k = k0 + c;
+      int k = k0 + c;
       float va = 0.f;
-      if (m < M && k < K) {
-        const int ix = m % d.IW, t1 = m / d.IW;
-        const int iy = t1 % d.IH, b = t1 / d.IH;
-        const int kx = k % d.KW, t2 = k / d.KW;
-        const int ky = t2 % d.KH, oc = t2 / d.KH;
-        const int ry = iy - ky, rx = ix - kx;
+      if (m_my < M && k < K) {
+        const int ry = iy0 - ky, rx = ix0 - kx;
         if (ry >= 0 && rx >= 0 && ry % d.S == 0 && rx % d.S == 0) {
           const int oy = ry / d.S, ox = rx / d.S;
           if (oy < d.OH && ox < d.OW) {
-            int64_t idx = (((int64_t)b * d.OC + oc) * d.OH + oy) * d.OW + ox;
+            int64_t idx = (((int64_t)b0 * d.OC + oc) * d.OH + oy) * d.OW + ox;
             va = dy[idx];
             if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
           }
@@ -213,6 +258,7 @@ void conv_dgrad_kernel(const float* __restrict__ dy,
       int n = bn0 + row;
       float vb = (n < d.IC && k < K) ? wt[(int64_t)n * K + k] : 0.f;
       lds_put<BF16>(ws, row, c, vb);
+      if (++kx == d.KW) { kx = 0; if (++ky == d.KH) { ky = 0; ++oc; } }
     }
     __syncthreads();
     mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
@@ -247,8 +293,8 @@ __global__ __launch_bounds__(256)
 void conv_wgrad_kernel(const float* __restrict__ dy,
                        const float* __restrict__ ymask,
                        const float* __restrict__ x,
-                       float* __restrict__ dw, float* __restrict__ db,
-                       ConvDims d) {
+                       float* __restrict__ part,   // [S][OC*K + OC]
+                       ConvDims d, int m_chunk) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -258,6 +304,10 @@ void conv_wgrad_kernel(const float* __restrict__ dy,
   const int bk0 = blockIdx.y * TB;   // k cols
   const int M = d.B * d.OH * d.OW;
   const int K = d.IC * d.KH * d.KW;
+  const int m_lo = blockIdx.z * m_chunk;
+  const int m_hi = min(M, m_lo + m_chunk);
+  float* dw_p = part + (int64_t)blockIdx.z * ((int64_t)d.OC * K + d.OC);
+  float* db_p = dw_p + (int64_t)d.OC * K;
   constexpr int BK = BF16 ? BKB : BKF;
   constexpr int EL = BF16 ? 16 : 4;
   constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
@@ -268,38 +318,45 @@ void conv_wgrad_kernel(const float* __restrict__ dy,
 
   const int row = tid & 63;
   const int c00 = (tid >> 6) * EL;
-  for (int i0 = 0; i0 < M; i0 += BK) {
-    // A tile: as[oc][i] = dYeff[i0+i, bn0+oc]
+  const int oc_my = bn0 + row;       // A-tile row (fixed per thread)
+  const int k_my = bk0 + row;        // B-tile row (fixed per thread)
+  const KDec kd = kdec(k_my < K ? k_my : 0, d);
+  for (int i0 = m_lo; i0 < m_hi; i0 += BK) {
+    MDec md = mdec((i0 + c00) < M ? (i0 + c00) : 0, d);
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
       int i = c00 + e;
-      int m = i0 + i, oc = bn0 + row;
+      int m = i0 + i;
+      // A tile: as[oc][i] = dYeff[m, oc_my]
       float va = 0.f;
-      if (m < M && oc < d.OC) {
-        const int ox = m % d.OW, t1 = m / d.OW;
-        const int oy = t1 % d.OH, b = t1 / d.OH;
-        int64_t idx = (((int64_t)b * d.OC + oc) * d.OH + oy) * d.OW + ox;
+      if (m < m_hi && oc_my < d.OC) {
+        int64_t idx = (((int64_t)md.b * d.OC + oc_my) * d.OH + md.oy)
+                          * d.OW + md.ox;
         va = dy[idx];
         if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
       }
       lds_put<BF16>(smem, row, i, va);
-      // B tile: bs[k][i] = im2col[i0+i, bk0+k]
-      int kcol = bk0 + row;
-      float vb = (m < M && kcol < K) ? im2col_at(x, d, m, kcol) : 0.f;
+      // B tile: bs[k][i] = im2col[m, k_my]
+      float vb = 0.f;
+      if (m < m_hi && k_my < K) {
+        int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
+        vb = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
+      }
       lds_put<BF16>(smem + LBYTES, row, i, vb);
+      minc(md, d);
     }
     __syncthreads();
     mma_tiles<BF16>(smem, smem + LBYTES, acc, lane, wrow, wcol);
-    if (db && blockIdx.y == 0 && tid < 64) {
-      float s = 0.f;
+    if (blockIdx.y == 0 && tid < 64) {
+      float sm = 0.f;
       if constexpr (BF16) {
         const __bf16* as = (const __bf16*)smem;
-        for (int i = 0; i < BKB; ++i) s += (float)as[tid * LDB + i];
+        for (int i = 0; i < BKB; ++i) sm += (float)as[tid * LDB + i];
       } else {
         const float* as = (const float*)smem;
-        for (int i = 0; i < BKF; ++i) s += as[tid * LDF + i];
+        for (int i = 0; i < BKF; ++i) sm += as[tid * LDF + i];
       }
-      dbs[tid] += s;
+      dbs[tid] += sm;
     }
     __syncthreads();
   }
@@ -313,10 +370,26 @@ void conv_wgrad_kernel(const float* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         int oc = bn0 + wrow + mi * 16 + crow + r;
         int k = bk0 + wcol + ni * 16 + ccol;
-        if (oc < d.OC && k < K) dw[(int64_t)oc * K + k] = acc[mi][ni][r];
+        if (oc < d.OC && k < K) dw_p[(int64_t)oc * K + k] = acc[mi][ni][r];
       }
-  if (db && blockIdx.y == 0 && tid < 64 && bn0 + tid < d.OC)
-    db[bn0 + tid] = dbs[tid];
+  if (blockIdx.y == 0 && tid < 64 && bn0 + tid < d.OC)
+    db_p[bn0 + tid] = dbs[tid];
+}
+
+// deterministic slab combine: dw[i] = sum_z part[z][i]; db likewise
+__global__ __launch_bounds__(256)
+void wgrad_combine_kernel(const float* __restrict__ part,
+                          float* __restrict__ dw, float* __restrict__ db,
+                          int64_t dw_n, int64_t oc_n, int n_slabs) {
+  const int64_t total = dw_n + oc_n;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float s = 0.f;
+    for (int z = 0; z < n_slabs; ++z) s += part[(int64_t)z * total + i];
+    if (i < dw_n) dw[i] = s;
+    else if (db) db[i - dw_n] = s;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -392,7 +465,16 @@ std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
   auto dw = torch::empty_like(w);
   auto db = torch::empty({d.OC}, w.options());
   const int K = d.IC * d.KH * d.KW;
-  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB);
+  const int M = d.B * d.OH * d.OW;
+  const int tiles = ((d.OC + TB - 1) / TB) * ((K + TB - 1) / TB);
+  // split the batch reduction so ~256+ blocks are in flight
+  const int BKc = 64;
+  int max_split = (M + BKc - 1) / BKc;
+  int split = std::max(1, std::min(max_split, (512 + tiles - 1) / tiles));
+  int m_chunk = ((M + split - 1) / split + BKc - 1) / BKc * BKc;
+  split = (M + m_chunk - 1) / m_chunk;
+  auto part = torch::empty({split, (int64_t)d.OC * K + d.OC}, w.options());
+  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB, split);
   const bool mask = ymask.has_value();
   const float* mp = mask ? ymask->data_ptr<float>() : nullptr;
   const bool bf16 = *g_bf16_flag2;
@@ -400,13 +482,18 @@ std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
     hipLaunchKernelGGL((conv_wgrad_kernel<decltype(b16)::value,
                                           decltype(mk)::value>),
                        grid, dim3(256), 0, stream(), dy.data_ptr<float>(),
-                       mp, x.data_ptr<float>(), dw.data_ptr<float>(),
-                       db.data_ptr<float>(), d);
+                       mp, x.data_ptr<float>(), part.data_ptr<float>(), d,
+                       m_chunk);
   };
   if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }
   else      { if (mask) L(std::false_type{}, std::true_type{});
               else L(std::false_type{}, std::false_type{}); }
+  int64_t dw_n = (int64_t)d.OC * K;
+  int blocks = (int)std::min<int64_t>((dw_n + d.OC + 255) / 256, 512);
+  hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
+                     stream(), part.data_ptr<float>(), dw.data_ptr<float>(),
+                     db.data_ptr<float>(), dw_n, (int64_t)d.OC, split);
   return {dw, db};
 }
 

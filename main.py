@@ -46,6 +46,8 @@ def load_session(run_id: str, device: torch.device):
     q_opt = FlatAdam(critic)
     q_opt.load_state_dict(auxiliaries["q_opt"])
     start_epoch = auxiliaries["epoch"]
+    # learned entropy temperature, if it was saved (extension)
+    pi_opt._resume_log_alpha = auxiliaries.get("log_alpha")
 
     # not SAC-constructor params (the reference forgets buffer_size and
     # crashes on resume when it was logged — fixed here)
@@ -184,6 +186,10 @@ def main():
 
     sac = SAC(**params, learn_alpha=args.learn_alpha,
               reference_pi_loss=args.reference_pi_loss)
+    resumed_la = getattr(pi_opt, "_resume_log_alpha", None)
+    if args.learn_alpha and resumed_la is not None:
+        import math
+        sac.alpha = math.exp(float(resumed_la))
     if args.normalize_states:
         from torch_actor_critic_amd.utils.normalizer import (
             WelfordVarianceEstimate)

@@ -124,3 +124,56 @@ def test_dp_ranks_stay_identical():
         p.join(timeout=30)
         assert p.exitcode == 0
     np.testing.assert_allclose(out[0], out[1], atol=1e-7)
+
+
+def _worker_sac_train(rank, world, port, q):
+    """Full SAC.train with 2 DP ranks (gloo): ranks must end with
+    IDENTICAL parameters (sync at start + grad all-reduce each update)."""
+    _dist_env(rank, world, port)
+    import numpy as np
+    from buffer.replay_buffer import ReplayBuffer
+    from networks.linear import Actor, DoubleCritic
+    from sac.algorithm import SAC
+    from torch_actor_critic_amd import envs
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.parallel import comm
+    comm.init_distributed(backend="gloo")
+
+    torch.manual_seed(50 + rank)   # different init; sync must fix it
+    env = envs.make("Pendulum-v1")
+    env.seed(rank)
+    actor = Actor(3, 1, [16, 16], act_limit=2.0)
+    critic = DoubleCritic(3, 1, [16, 16])
+    buf = ReplayBuffer(2000, 3, 1)
+    pi_opt, q_opt = FlatAdam(actor), FlatAdam(critic)
+    sac = SAC(alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=16, start_steps=60, steps_per_epoch=200,
+              max_ep_len=100, update_after=60, update_every=20,
+              save_every=100)
+    m = sac.train(0, env, actor, critic, buf, pi_opt, q_opt,
+                  render=False, logging=False)
+    q.put((rank, pi_opt.fp.flat.clone().numpy(),
+           q_opt.fp.flat.clone().numpy(), m["loss_q"]))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp_sac_train_end_to_end():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29533
+    procs = [ctx.Process(target=_worker_sac_train, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(2):
+        rank, aflat, cflat, lq = q.get(timeout=280)
+        out[rank] = (aflat, cflat, lq)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    np.testing.assert_allclose(out[0][0], out[1][0], atol=1e-6)
+    np.testing.assert_allclose(out[0][1], out[1][1], atol=1e-6)
+    assert np.isfinite(out[0][2]) and out[0][2] != 0.0

@@ -171,8 +171,10 @@ def main():
 
     act_graph = None
     if on_gpu and not visual:
-        from torch_actor_critic_amd.algo.act import ActGraph, WindowedStore
-        act_graph = ActGraph(actor, obs_dim, act_dim, device)
+        from torch_actor_critic_amd.algo.act import (WindowedStore,
+                                                     make_act_path)
+        act_graph = make_act_path(actor, obs_dim, act_dim, device,
+                                  4321 + 7919 * rank)
         wstore = WindowedStore(buffer, args.update_window)
 
     do_act = args.mode in ("full", "acting")

@@ -287,3 +287,36 @@ def test_sac_train_on_gpu_learns_pendulum():
             ep += r
         eval_rets.append(ep)
     assert float(np.mean(eval_rets)) > -700.0, eval_rets
+
+
+def test_act_kernel_matches_actor():
+    """The one-launch act kernel equals the eager actor when noise is
+    silenced (log_std bias at the clamp floor), and produces bounded,
+    varying actions stochastically."""
+    from torch_actor_critic_amd.algo.act import ActKernel
+    from torch_actor_critic_amd.models.mlp import Actor
+
+    torch.manual_seed(21)
+    device = torch.device(DEV)
+    actor = Actor(17, 6, [256, 256], act_limit=1.5).to(device)
+    ak = ActKernel(actor, 17, 6, device, philox_seed=3)
+
+    state = np.random.default_rng(0).standard_normal(17).astype(np.float32)
+
+    # silence the noise: log_std ~ -30 -> clamped to -20 -> std ~ 2e-9
+    with torch.no_grad():
+        actor.log_std_layer.bias.fill_(-30.0)
+        actor.log_std_layer.weight.zero_()
+    a_kernel = ak.act(state)
+    with torch.no_grad():
+        a_ref, _ = actor(torch.as_tensor(state, device=device),
+                         deterministic=True, with_logprob=False)
+    np.testing.assert_allclose(a_kernel, a_ref.cpu().numpy(), atol=1e-4)
+
+    # stochastic: bounded and varying draw to draw
+    with torch.no_grad():
+        actor.log_std_layer.bias.fill_(-1.0)
+    a1 = ak.act(state)
+    a2 = ak.act(state)
+    assert np.all(np.abs(a1) <= 1.5 + 1e-5)
+    assert not np.allclose(a1, a2)

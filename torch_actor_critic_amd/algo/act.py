@@ -18,6 +18,60 @@ import numpy as np
 import torch
 
 
+class ActKernel:
+    """Single-kernel acting path: the whole stochastic actor forward for
+    one state runs as ONE launch (ops/csrc/fused.hip::act_kernel) — no
+    graph replay, no per-layer launches, self-bumped Philox noise."""
+
+    def __init__(self, actor, obs_dim: int, act_dim: int,
+                 device: torch.device, philox_seed: int = 0):
+        from ..models.mlp import Actor as MlpActor
+        from ..ops import require_extension, functional as Fo
+        if type(actor) is not MlpActor:
+            raise TypeError("ActKernel needs the plain MLP Actor")
+        if any(l.out_features > 256 for l in actor.layers) or act_dim > 64:
+            raise ValueError("shape outside act-kernel limits")
+        self.ext = require_extension()
+        self.ws = [l.weight for l in actor.layers]
+        self.bs = [l.bias for l in actor.layers]
+        self.wmu = actor.mu_layer.weight
+        self.bmu = actor.mu_layer.bias
+        self.wls = actor.log_std_layer.weight
+        self.bls = actor.log_std_layer.bias
+        self.act_limit = float(actor.act_limit)
+        self.lo = float(actor.log_min_std)
+        self.hi = float(actor.log_max_std)
+        self.seed = philox_seed ^ 0x5ACB
+        self.ctr = torch.zeros(1, dtype=torch.int64, device=device)
+        self.obs_in = torch.zeros(obs_dim, device=device)
+        self.obs_pin = torch.zeros(obs_dim, pin_memory=True)
+        self.act_out = torch.zeros(act_dim, device=device)
+        self.act_pin = torch.zeros(act_dim, pin_memory=True)
+        self.ev = torch.cuda.Event()
+
+    def act(self, state: np.ndarray) -> np.ndarray:
+        self.obs_pin.copy_(torch.from_numpy(np.asarray(state,
+                                                       dtype=np.float32)))
+        self.obs_in.copy_(self.obs_pin, non_blocking=True)
+        self.ext.act_step(self.obs_in, self.ws, self.bs, self.wmu,
+                          self.bmu, self.wls, self.bls, self.act_out,
+                          self.ctr, self.seed, self.act_limit, self.lo,
+                          self.hi)
+        self.act_pin.copy_(self.act_out, non_blocking=True)
+        self.ev.record()
+        self.ev.synchronize()
+        return self.act_pin.numpy().copy()
+
+
+def make_act_path(actor, obs_dim, act_dim, device, philox_seed: int = 0):
+    """Best available single-state acting path: the one-launch act
+    kernel when the actor fits its limits, else the captured act-graph."""
+    try:
+        return ActKernel(actor, obs_dim, act_dim, device, philox_seed)
+    except Exception:  # noqa: BLE001
+        return ActGraph(actor, obs_dim, act_dim, device)
+
+
 class ActGraph:
     """hipGraph-captured single-state stochastic actor forward."""
 

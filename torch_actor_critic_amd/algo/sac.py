@@ -345,9 +345,10 @@ class SAC:
         if (device.type == "cuda" and self.use_graph
                 and hasattr(buffer, "obs_dim")):
             try:
-                from .act import ActGraph, WindowedStore
-                act_graph = ActGraph(actor, buffer.obs_dim, buffer.act_dim,
-                                     device)
+                from .act import WindowedStore, make_act_path
+                act_graph = make_act_path(actor, buffer.obs_dim,
+                                          buffer.act_dim, device,
+                                          10000 * comm.proc_id())
                 wstore = WindowedStore(buffer, self.update_every)
             except Exception as e:  # noqa: BLE001
                 logger.warning("act-graph capture failed (%r); using eager "

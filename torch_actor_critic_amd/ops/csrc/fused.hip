@@ -790,6 +790,79 @@ void piloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Whole-policy single-state action kernel: the entire actor forward for
+// ONE state (the serial env-interaction path) in ONE launch — trunk
+// GEMVs through LDS, dual heads, clamp/exp, Philox noise (single block,
+// so the counter self-bumps: no predecessor kernel), tanh squash.
+// ---------------------------------------------------------------------------
+
+constexpr int ACT_MAXL = 4;
+constexpr int ACT_MAXW = 512;
+
+struct ActArgs {
+  const float* x;              // [O] device input state
+  const float* w[ACT_MAXL];    // trunk weights [H, K]
+  const float* b[ACT_MAXL];
+  int width[ACT_MAXL];
+  int n_layers, O, A;
+  const float* wmu; const float* bmu;
+  const float* wls; const float* bls;
+  float* action;               // [A]
+  int64_t* ctr;
+  uint64_t seed;
+  float act_limit, lo, hi;
+};
+
+__global__ __launch_bounds__(256)
+void act_kernel(ActArgs a) {
+  __shared__ __attribute__((aligned(16))) float buf[2][ACT_MAXW];
+  __shared__ unsigned long long ctr_s;
+  const int tid = threadIdx.x;
+  if (tid == 0) ctr_s = (unsigned long long)(++a.ctr[0]);
+  for (int k = tid; k < a.O; k += blockDim.x) buf[0][k] = a.x[k];
+  __syncthreads();
+
+  int cur = 0, K = a.O;
+  for (int L = 0; L < a.n_layers; ++L) {
+    const int H = a.width[L];
+    float acc = 0.f;
+    if (tid < H) {
+      acc = a.b[L][tid];
+      const float* wr = a.w[L] + (int64_t)tid * K;
+      int k = 0;
+      for (; k + 4 <= K; k += 4) {
+        acc += wr[k] * buf[cur][k] + wr[k+1] * buf[cur][k+1]
+             + wr[k+2] * buf[cur][k+2] + wr[k+3] * buf[cur][k+3];
+      }
+      for (; k < K; ++k) acc += wr[k] * buf[cur][k];
+    }
+    __syncthreads();
+    if (tid < H) buf[1 - cur][tid] = fmaxf(acc, 0.f);
+    __syncthreads();
+    cur ^= 1;
+    K = H;
+  }
+
+  // heads + tanh-Gaussian sample (stochastic acting path)
+  if (tid < a.A) {
+    float mu = a.bmu[tid], ls = a.bls[tid];
+    const float* wm = a.wmu + (int64_t)tid * K;
+    const float* wl = a.wls + (int64_t)tid * K;
+    for (int k = 0; k < K; ++k) {
+      float h = buf[cur][k];
+      mu += wm[k] * h;
+      ls += wl[k] * h;
+    }
+    ls = fminf(fmaxf(ls, a.lo), a.hi);
+    P4 r = philox_(a.seed ^ 0x517cc1b727220a95ull, ctr_s, (uint64_t)tid);
+    float u0 = (r.x + 1.f) * 2.3283064365386963e-10f;
+    float u1 = (r.y + 1.f) * 2.3283064365386963e-10f;
+    float eps = sqrtf(-2.f * logf(u0)) * __cosf(6.283185307179586f * u1);
+    a.action[tid] = tanhf(mu + expf(ls) * eps) * a.act_limit;
+  }
+}
+
 // learned entropy temperature: one-thread Adam on log_alpha
 // (loss = -log_alpha * (mean_logp + target_entropy))
 __global__ void alpha_update_kernel(float* __restrict__ log_alpha,
@@ -1080,6 +1153,39 @@ torch::Tensor tg_eps(int64_t ctr_val, int64_t seed, int64_t R, int64_t A,
   return out;
 }
 
+void act_step(torch::Tensor x, std::vector<torch::Tensor> ws,
+              std::vector<torch::Tensor> bs, torch::Tensor wmu,
+              torch::Tensor bmu, torch::Tensor wls, torch::Tensor bls,
+              torch::Tensor action, torch::Tensor ctr, int64_t seed,
+              double act_limit, double lo, double hi) {
+  ActArgs a{};
+  a.n_layers = (int)ws.size();
+  TORCH_CHECK(a.n_layers <= ACT_MAXL);
+  a.x = x.data_ptr<float>();
+  a.O = (int)x.numel();
+  a.A = (int)action.numel();
+  int maxw = a.O;
+  for (int L = 0; L < a.n_layers; ++L) {
+    a.w[L] = ws[L].data_ptr<float>();
+    a.b[L] = bs[L].data_ptr<float>();
+    a.width[L] = (int)ws[L].size(0);
+    TORCH_CHECK(a.width[L] <= 256, "act kernel: trunk width > 256");
+    maxw = std::max(maxw, a.width[L]);
+  }
+  TORCH_CHECK(maxw <= ACT_MAXW && a.A <= 256);
+  a.wmu = wmu.data_ptr<float>();
+  a.bmu = bmu.data_ptr<float>();
+  a.wls = wls.data_ptr<float>();
+  a.bls = bls.data_ptr<float>();
+  a.action = action.data_ptr<float>();
+  a.ctr = ctr.data_ptr<int64_t>();
+  a.seed = (uint64_t)seed;
+  a.act_limit = (float)act_limit;
+  a.lo = (float)lo;
+  a.hi = (float)hi;
+  hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(), a);
+}
+
 void alpha_update(torch::Tensor log_alpha, torch::Tensor alpha_dev,
                   torch::Tensor m, torch::Tensor v, torch::Tensor step,
                   torch::Tensor mean_logp, double target_entropy, double lr) {
@@ -1105,4 +1211,5 @@ void register_fused(pybind11::module_& m) {
   m.def("alpha_update", &fused::alpha_update);
   m.def("tg_eps", &fused::tg_eps);
   m.def("adam_t", &fused::adam_t);
+  m.def("act_step", &fused::act_step);
 }

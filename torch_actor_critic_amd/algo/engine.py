@@ -133,6 +133,9 @@ class FusedSACEngine:
             return [(w.data_ptr() - base) // 4 for w in srcs]
         self._c_offs = offs(q_opt, self._c_tr_src)
         self._a_offs = offs(pi_opt, self._a_tr_src)
+        # final-layer dgrad fusion into the loss kernels (B-limited by
+        # the loss kernels' LDS dq stash; needs >= 2 critic layers)
+        self._loss_fuse = B <= 1024 and len(self.c_w) >= 2
 
         # whole-MLP fused forward feasibility (LDS budget)
         from ..ops import functional as Fo
@@ -273,10 +276,17 @@ class FusedSACEngine:
         nL = len(self.c_w)
         q = [self.c_act[z][nL - 1] for z in range(2)]
         qt = [self.t_act[z][nL - 1] for z in range(2)]
+        # final-layer dgrad (K=1 outer product) fuses into the loss kernel
+        fuse = self._loss_fuse
         ext.qloss2(q[0], q[1], qt[0], qt[1], self.logp[B:], self.rew,
                    self.done, self.alpha_dev, self.alpha_host,
                    self.loss_q_acc, self.dq[0], self.dq[1], B,
-                   self.sac.gamma, self.sac.reward_scale)
+                   self.sac.gamma, self.sac.reward_scale,
+                   self.cwt[0][nL - 1] if fuse else None,
+                   self.cwt[1][nL - 1] if fuse else None,
+                   self.dc[0][nL - 2] if fuse else None,
+                   self.dc[1][nL - 2] if fuse else None,
+                   self.c_w[nL - 2] if fuse else 0)
 
         # critic backward (wgrad into flat grads; dgrad via cached W^T)
         d = self.dq
@@ -296,11 +306,14 @@ class FusedSACEngine:
                        B, self.c_w[i], (self.c_w[i - 1] if i > 0 else OC),
                        self.c_w[i], ldx, xoff)
             if i > 0:
-                self._mg(d, [self.cwt[z][i] for z in range(2)],
-                         [None, None],
-                         [self.dc[z][i - 1] for z in range(2)],
-                         masks, B, self.c_w[i - 1], self.c_w[i],
-                         self.c_w[i], self.c_w[i - 1], False)
+                if i == nL - 1 and fuse:
+                    pass  # dy2 already produced by the fused loss kernel
+                else:
+                    self._mg(d, [self.cwt[z][i] for z in range(2)],
+                             [None, None],
+                             [self.dc[z][i - 1] for z in range(2)],
+                             masks, B, self.c_w[i - 1], self.c_w[i],
+                             self.c_w[i], self.c_w[i - 1], False)
                 d = [self.dc[z][i - 1] for z in range(2)]
 
     def _phase_policy(self):
@@ -317,15 +330,24 @@ class FusedSACEngine:
         self._critic_fwd(self.XC2, 0, self.cw, self.p_act, OC)
         nL = len(self.c_w)
         qp = [self.p_act[z][nL - 1] for z in range(2)]
+        fuse = self._loss_fuse
         ext.piloss2(qp[0], qp[1], self.logp[:B], self.alpha_dev,
                     self.alpha_host, self.loss_pi_acc, self.mean_logp,
-                    self.dqp[0], self.dqp[1], B)
+                    self.dqp[0], self.dqp[1], B,
+                    self.cwt[0][nL - 1] if fuse else None,
+                    self.cwt[1][nL - 1] if fuse else None,
+                    self.dcp[0][nL - 2] if fuse else None,
+                    self.dcp[1][nL - 2] if fuse else None,
+                    self.c_w[nL - 2] if fuse else 0)
 
         # critic dgrad chain only (frozen critic)
         d = self.dqp
         for i in range(nL - 1, 0, -1):
             masks = [self.p_act[z][i] if i + 1 < nL else None
                      for z in range(2)]
+            if i == nL - 1 and fuse:
+                d = [self.dcp[z][i - 1] for z in range(2)]
+                continue
             self._mg(d, [self.cwt[z][i] for z in range(2)], [None, None],
                      [self.dcp[z][i - 1] for z in range(2)], masks,
                      B, self.c_w[i - 1], self.c_w[i], self.c_w[i],

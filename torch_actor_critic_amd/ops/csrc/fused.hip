@@ -723,6 +723,11 @@ void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
 // Single-block losses (deterministic reductions, device-alpha aware)
 // ---------------------------------------------------------------------------
 
+// qloss: Bellman backup + twin-min MSE + gradient seeds + (optionally)
+// the K=1 dgrad of the final critic layer fused in: dy2_z = dq_z x wt3_z
+// (outer product; dq staged in LDS so no extra launch is needed).
+constexpr int LOSS_MAXB = 1024;
+
 __global__ __launch_bounds__(256)
 void qloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
                    const float* __restrict__ q1t, const float* __restrict__ q2t,
@@ -731,17 +736,25 @@ void qloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
                    const float* __restrict__ alpha_dev, float alpha_host,
                    float* __restrict__ loss_acc,
                    float* __restrict__ dq1, float* __restrict__ dq2,
-                   int B, float gamma, float scale) {
+                   int B, float gamma, float scale,
+                   const float* __restrict__ wt3_0,
+                   const float* __restrict__ wt3_1,
+                   float* __restrict__ dy2_0, float* __restrict__ dy2_1,
+                   int h2) {
   __shared__ float red[4];
+  __shared__ float dqs[2][LOSS_MAXB];
   const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
+  const bool fuse = dy2_0 != nullptr;
   float acc = 0.f;
   for (int i = threadIdx.x; i < B; i += blockDim.x) {
     float backup = scale * rew[i] + gamma * (1.f - done[i]) *
                        (fminf(q1t[i], q2t[i]) - alpha * logp_next[i]);
     float e1 = q1[i] - backup, e2 = q2[i] - backup;
     acc += e1 * e1 + e2 * e2;
-    dq1[i] = 2.f * e1 / B;
-    dq2[i] = 2.f * e2 / B;
+    float g1 = 2.f * e1 / B, g2 = 2.f * e2 / B;
+    dq1[i] = g1;
+    dq2[i] = g2;
+    if (fuse) { dqs[0][i] = g1; dqs[1][i] = g2; }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
@@ -749,6 +762,14 @@ void qloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
   __syncthreads();
   if (threadIdx.x == 0)
     loss_acc[0] += (red[0] + red[1] + red[2] + red[3]) / B;
+  if (fuse) {
+    const int64_t total = (int64_t)B * h2;
+    for (int64_t idx = threadIdx.x; idx < total; idx += blockDim.x) {
+      int i = (int)(idx / h2), j = (int)(idx % h2);
+      dy2_0[idx] = dqs[0][i] * wt3_0[j];
+      dy2_1[idx] = dqs[1][i] * wt3_1[j];
+    }
+  }
 }
 
 __global__ __launch_bounds__(256)
@@ -758,9 +779,15 @@ void piloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
                     float* __restrict__ loss_acc,
                     float* __restrict__ mean_logp,
                     float* __restrict__ dq1, float* __restrict__ dq2,
-                    int B) {
+                    int B,
+                    const float* __restrict__ wt3_0,
+                    const float* __restrict__ wt3_1,
+                    float* __restrict__ dy2_0, float* __restrict__ dy2_1,
+                    int h2) {
   __shared__ float red[4], redl[4];
+  __shared__ float dqs[2][LOSS_MAXB];
   const float alpha = alpha_dev ? alpha_dev[0] : alpha_host;
+  const bool fuse = dy2_0 != nullptr;
   float acc = 0.f, laux = 0.f;
   for (int i = threadIdx.x; i < B; i += blockDim.x) {
     float a = q1[i], b = q2[i];
@@ -772,6 +799,7 @@ void piloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
     else            { g1 = -0.5f; g2 = -0.5f; }
     dq1[i] = g1 / B;
     dq2[i] = g2 / B;
+    if (fuse) { dqs[0][i] = g1 / B; dqs[1][i] = g2 / B; }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
@@ -787,6 +815,14 @@ void piloss2_kernel(const float* __restrict__ q1, const float* __restrict__ q2,
     loss_acc[0] += (red[0] + red[1] + red[2] + red[3]) / B;
     if (mean_logp)
       mean_logp[0] = (redl[0] + redl[1] + redl[2] + redl[3]) / B;
+  }
+  if (fuse) {
+    const int64_t total = (int64_t)B * h2;
+    for (int64_t idx = threadIdx.x; idx < total; idx += blockDim.x) {
+      int i = (int)(idx / h2), j = (int)(idx % h2);
+      dy2_0[idx] = dqs[0][i] * wt3_0[j];
+      dy2_1[idx] = dqs[1][i] * wt3_1[j];
+    }
   }
 }
 
@@ -1098,7 +1134,13 @@ void qloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor q1t,
             torch::Tensor q2t, torch::Tensor logp_next, torch::Tensor rew,
             torch::Tensor done, c10::optional<torch::Tensor> alpha_dev,
             double alpha_host, torch::Tensor loss_acc, torch::Tensor dq1,
-            torch::Tensor dq2, int64_t B, double gamma, double scale) {
+            torch::Tensor dq2, int64_t B, double gamma, double scale,
+            c10::optional<torch::Tensor> wt3_0,
+            c10::optional<torch::Tensor> wt3_1,
+            c10::optional<torch::Tensor> dy2_0,
+            c10::optional<torch::Tensor> dy2_1, int64_t h2) {
+  const bool fuse = dy2_0.has_value();
+  TORCH_CHECK(!fuse || B <= LOSS_MAXB);
   hipLaunchKernelGGL(qloss2_kernel, dim3(1), dim3(256), 0, stream(),
                      q1.data_ptr<float>(), q2.data_ptr<float>(),
                      q1t.data_ptr<float>(), q2t.data_ptr<float>(),
@@ -1106,20 +1148,32 @@ void qloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor q1t,
                      done.data_ptr<float>(), fptr(alpha_dev),
                      (float)alpha_host, loss_acc.data_ptr<float>(),
                      dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B,
-                     (float)gamma, (float)scale);
+                     (float)gamma, (float)scale,
+                     fptr(wt3_0), fptr(wt3_1),
+                     fuse ? dy2_0->data_ptr<float>() : nullptr,
+                     fuse ? dy2_1->data_ptr<float>() : nullptr, (int)h2);
 }
 
 void piloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor logp,
              c10::optional<torch::Tensor> alpha_dev, double alpha_host,
              torch::Tensor loss_acc, c10::optional<torch::Tensor> mean_logp,
-             torch::Tensor dq1, torch::Tensor dq2, int64_t B) {
+             torch::Tensor dq1, torch::Tensor dq2, int64_t B,
+             c10::optional<torch::Tensor> wt3_0,
+             c10::optional<torch::Tensor> wt3_1,
+             c10::optional<torch::Tensor> dy2_0,
+             c10::optional<torch::Tensor> dy2_1, int64_t h2) {
+  const bool fuse = dy2_0.has_value();
+  TORCH_CHECK(!fuse || B <= LOSS_MAXB);
   hipLaunchKernelGGL(piloss2_kernel, dim3(1), dim3(256), 0, stream(),
                      q1.data_ptr<float>(), q2.data_ptr<float>(),
                      logp.data_ptr<float>(), fptr(alpha_dev),
                      (float)alpha_host, loss_acc.data_ptr<float>(),
                      mean_logp.has_value() ? mean_logp->data_ptr<float>()
                                            : nullptr,
-                     dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B);
+                     dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B,
+                     fptr(wt3_0), fptr(wt3_1),
+                     fuse ? dy2_0->data_ptr<float>() : nullptr,
+                     fuse ? dy2_1->data_ptr<float>() : nullptr, (int)h2);
 }
 
 void adam_t(torch::Tensor p, torch::Tensor g, torch::Tensor m,

@@ -134,8 +134,12 @@ class FusedSACEngine:
         self._c_offs = offs(q_opt, self._c_tr_src)
         self._a_offs = offs(pi_opt, self._a_tr_src)
         # final-layer dgrad fusion into the loss kernels (B-limited by
-        # the loss kernels' LDS dq stash; needs >= 2 critic layers)
-        self._loss_fuse = B <= 1024 and len(self.c_w) >= 2
+        # the loss kernels' LDS dq stash; needs >= 2 critic layers);
+        # TAC_AMD_LOSS_FUSE={0,1} overrides for A/B measurement
+        import os as _os
+        _lf = _os.environ.get("TAC_AMD_LOSS_FUSE")
+        self._loss_fuse = (B <= 1024 and len(self.c_w) >= 2
+                           if _lf is None else _lf == "1")
 
         # whole-MLP fused forward feasibility (LDS budget)
         from ..ops import functional as Fo

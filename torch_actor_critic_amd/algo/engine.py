@@ -133,13 +133,13 @@ class FusedSACEngine:
             return [(w.data_ptr() - base) // 4 for w in srcs]
         self._c_offs = offs(q_opt, self._c_tr_src)
         self._a_offs = offs(pi_opt, self._a_tr_src)
-        # final-layer dgrad fusion into the loss kernels (B-limited by
-        # the loss kernels' LDS dq stash; needs >= 2 critic layers);
-        # TAC_AMD_LOSS_FUSE={0,1} overrides for A/B measurement
+        # Final-layer dgrad fusion into the loss kernels: measured
+        # SLOWER (A/B 4344 vs 4613 upd/s at batch 64) — the single-block
+        # outer-product write loses to the 8-block GEMM it replaces.
+        # Kept for study via TAC_AMD_LOSS_FUSE=1; default off.
         import os as _os
-        _lf = _os.environ.get("TAC_AMD_LOSS_FUSE")
-        self._loss_fuse = (B <= 1024 and len(self.c_w) >= 2
-                           if _lf is None else _lf == "1")
+        self._loss_fuse = (_os.environ.get("TAC_AMD_LOSS_FUSE") == "1"
+                           and B <= 1024 and len(self.c_w) >= 2)
 
         # whole-MLP fused forward feasibility (LDS budget)
         from ..ops import functional as Fo

@@ -22,3 +22,10 @@ tensorboard:
 
 lint:
 	python -m flake8 torch_actor_critic_amd sac networks buffer environments tests main.py run_agent.py bench.py
+
+# Reference-parity convenience targets (reference Makefile:17-28)
+dvc:
+	dvc repro
+
+prefect:
+	prefect server start

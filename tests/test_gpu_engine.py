@@ -320,3 +320,24 @@ def test_act_kernel_matches_actor():
     a2 = ak.act(state)
     assert np.all(np.abs(a1) <= 1.5 + 1e-5)
     assert not np.allclose(a1, a2)
+
+
+def test_main_cli_on_gpu(tmp_path):
+    """Driver-style usage: short real training run through main.py on
+    the GPU fast path, checkpoint written, resumable."""
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "main.py"),
+         "--environment", "HalfCheetah-v4", "--epochs", "10",
+         "--steps-per-epoch", "150", "--batch-size", "64",
+         "--buffer-size", "20000"],
+        cwd=tmp_path, timeout=420, capture_output=True, text=True,
+        env={**os.environ, "PYTHONPATH": repo})
+    assert r.returncode == 0, r.stderr[-2000:]
+    runs = os.listdir(tmp_path / "mlruns" / "0")
+    assert len(runs) == 1
+    art = tmp_path / "mlruns" / "0" / runs[0] / "artifacts"
+    assert (art / "actor" / "data" / "model.pth").exists()

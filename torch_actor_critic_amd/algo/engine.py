@@ -234,7 +234,10 @@ class FusedSACEngine:
         ext = self.ext
         B, O, A, OC = self.B, self.O, self.A, self.OC
         buf = self.buffer
-        ext.bump_counter(self.ctr)
+        # one launch bumps the replay/noise counter AND both Adam step
+        # counters (their kernels read the committed values later on the
+        # same stream)
+        ext.bump3(self.ctr, self.q_opt.step_t, self.pi_opt.step_t)
         if self.sample:
             ext.gather2(buf.state, buf.actions, buf.rewards, buf.next_state,
                         buf.done, buf._size_dev, self.ctr, self.seed,
@@ -323,9 +326,8 @@ class FusedSACEngine:
     def _phase_policy(self):
         ext = self.ext
         B, O, A, OC = self.B, self.O, self.A, self.OC
-        # fused Adam + transposed-weight refresh (one bump + one kernel)
+        # fused Adam + transposed-weight refresh (step already bumped)
         qo = self.q_opt
-        ext.bump_counter(qo.step_t)
         ext.adam_t(qo.fp.flat, qo.fp.flat_grad, qo.m, qo.v, qo.step_t,
                    qo.lr, qo.betas[0], qo.betas[1], qo.eps,
                    qo.weight_decay, self._c_offs, self._c_tr_dst)
@@ -400,7 +402,6 @@ class FusedSACEngine:
     def _phase_finish(self):
         ext = self.ext
         po = self.pi_opt
-        ext.bump_counter(po.step_t)
         ext.adam_t(po.fp.flat, po.fp.flat_grad, po.m, po.v, po.step_t,
                    po.lr, po.betas[0], po.betas[1], po.eps,
                    po.weight_decay, self._a_offs, self._a_tr_dst)

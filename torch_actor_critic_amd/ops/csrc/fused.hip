@@ -673,6 +673,14 @@ void tg_fwd2_kernel(const float* __restrict__ hl,  // [R, 2A]: mu | log_std
   if (a == 0) logp[b] = acc;
 }
 
+// one launch bumping several device counters (replay ctr + both Adam
+// step counters at update start — saves two 1-thread launches)
+__global__ void bump3_kernel(int64_t* a, int64_t* b, int64_t* c) {
+  if (a) ++a[0];
+  if (b) ++b[0];
+  if (c) ++c[0];
+}
+
 // Debug/test helper: reproduce tg_fwd2's internal noise exactly so the
 // eager fp32 reference can be driven with identical eps.
 __global__ __launch_bounds__(64)
@@ -1240,6 +1248,14 @@ void act_step(torch::Tensor x, std::vector<torch::Tensor> ws,
   hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(), a);
 }
 
+void bump3(torch::Tensor a, c10::optional<torch::Tensor> b,
+           c10::optional<torch::Tensor> c) {
+  hipLaunchKernelGGL(bump3_kernel, dim3(1), dim3(1), 0, stream(),
+                     a.data_ptr<int64_t>(),
+                     b.has_value() ? b->data_ptr<int64_t>() : nullptr,
+                     c.has_value() ? c->data_ptr<int64_t>() : nullptr);
+}
+
 void alpha_update(torch::Tensor log_alpha, torch::Tensor alpha_dev,
                   torch::Tensor m, torch::Tensor v, torch::Tensor step,
                   torch::Tensor mean_logp, double target_entropy, double lr) {
@@ -1266,4 +1282,5 @@ void register_fused(pybind11::module_& m) {
   m.def("tg_eps", &fused::tg_eps);
   m.def("adam_t", &fused::adam_t);
   m.def("act_step", &fused::act_step);
+  m.def("bump3", &fused::bump3);
 }

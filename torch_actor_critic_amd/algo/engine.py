@@ -101,6 +101,7 @@ class FusedSACEngine:
         self.dcp = cbufs(B)       # pi-phase dgrad chain buffers
         self.dc = cbufs(B)        # q-phase dgrad chain buffers
         self.dxc = torch.zeros(B, OC, **f32)
+        self.dxc2 = torch.zeros(B, OC, **f32)
 
         # ---- module parameter views -----------------------------------
         def critic_layers(mod):
@@ -190,10 +191,11 @@ class FusedSACEngine:
     # ------------------------------------------------------------------
 
     def _mg(self, xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
-            xs2=None, ws2=None, masks2=None, K2=0, x_off=0, x2_off=0):
+            xs2=None, ws2=None, masks2=None, K2=0, x_off=0, x2_off=0,
+            x_offs=None):
         self.ext.mgemm(xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
                        xs2 or [], ws2 or [], masks2 or [], K2, x_off,
-                       x2_off)
+                       x2_off, x_offs or [])
 
     def _critic_fwd(self, x_src, x_off, weights, acts, lda,
                     need_hidden_acts=True):
@@ -274,11 +276,34 @@ class FusedSACEngine:
                     B, self.logp, self.prob, self.ctr, self.seed,
                     self.act_limit, self.lo, self.hi)
 
-        # target critic on (ns, a2) = XC rows B:
-        self._critic_fwd(self.XC, B * OC, self.tw, self.t_act, OC,
-                         need_hidden_acts=False)
-        # critic on (s, a) = XC rows :B
-        self._critic_fwd(self.XC, 0, self.cw, self.c_act, OC)
+        # target critic on (ns, a2) = XC rows B: AND the live critic on
+        # (s, a) = XC rows :B — same shapes, one 4-problem launch per layer
+        if not self.use_mlpf:
+            nLc = len(self.c_w)
+            xs4 = [self.XC] * 4
+            offs4 = [B * OC, B * OC, 0, 0]
+            k = OC
+            lda = OC
+            for i in range(nLc):
+                relu = i + 1 < nLc
+                self._mg(xs4,
+                         [self.tw[0][i][0], self.tw[1][i][0],
+                          self.cw[0][i][0], self.cw[1][i][0]],
+                         [self.tw[0][i][1], self.tw[1][i][1],
+                          self.cw[0][i][1], self.cw[1][i][1]],
+                         [self.t_act[0][i], self.t_act[1][i],
+                          self.c_act[0][i], self.c_act[1][i]],
+                         [None] * 4, B, self.c_w[i], k, lda, self.c_w[i],
+                         relu, x_offs=offs4)
+                xs4 = [self.t_act[0][i], self.t_act[1][i],
+                       self.c_act[0][i], self.c_act[1][i]]
+                offs4 = [0, 0, 0, 0]
+                k = self.c_w[i]
+                lda = k
+        else:
+            self._critic_fwd(self.XC, B * OC, self.tw, self.t_act, OC,
+                             need_hidden_acts=False)
+            self._critic_fwd(self.XC, 0, self.cw, self.c_act, OC)
 
         nL = len(self.c_w)
         q = [self.c_act[z][nL - 1] for z in range(2)]
@@ -359,17 +384,17 @@ class FusedSACEngine:
                      B, self.c_w[i - 1], self.c_w[i], self.c_w[i],
                      self.c_w[i - 1], False)
             d = [self.dcp[z][i - 1] for z in range(2)]
-        # layer 0: sum over the twin critics -> dxc
+        # layer 0: per-critic dxc slabs in one z=2 launch; tg_bwd2 sums
+        # them on read (measured faster than the serial two-pass sum2)
         masks0 = [self.p_act[z][0] for z in range(2)]
-        self._mg([d[0]], [self.cwt[0][0]], [None], [self.dxc], [masks0[0]],
-                 B, OC, self.c_w[0], self.c_w[0], OC, False,
-                 xs2=[d[1]], ws2=[self.cwt[1][0]], masks2=[masks0[1]],
-                 K2=self.c_w[0])
+        self._mg(d, [self.cwt[z][0] for z in range(2)], [None, None],
+                 [self.dxc, self.dxc2], masks0,
+                 B, OC, self.c_w[0], self.c_w[0], OC, False)
 
         # actor backward
-        ext.tg_bwd2(self.dxc, O, self.alpha_dev, self.alpha_host,
-                    self.hl, self.prob, self.dmu, self.dls, B,
-                    self.act_limit, self.lo, self.hi)
+        ext.tg_bwd2(self.dxc, O, self.dxc2, self.alpha_dev,
+                    self.alpha_host, self.hl, self.prob, self.dmu,
+                    self.dls, B, self.act_limit, self.lo, self.hi)
         h_last = self.a_hidden[-1]
         a_last = self.a_act[-1]
         (wm, bm), (wl, bl) = self.head_w

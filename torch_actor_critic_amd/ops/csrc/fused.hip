@@ -76,8 +76,10 @@ struct MProb {
   const float* x2; const float* w2; const float* mask2;
 };
 
+constexpr int MAXZ = 4;
+
 struct MGemm {
-  MProb p[2];
+  MProb p[MAXZ];
   int M, N, K, lda, ldy;
   int K2;  // reduction depth of the second operand pair (SUM2)
 };
@@ -701,6 +703,7 @@ void tg_eps_kernel(float* __restrict__ eps_out, uint64_t ctr_val,
 // in-kernel from the device alpha.
 __global__ __launch_bounds__(64)
 void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
+                    const float* __restrict__ dxc2,  // optional 2nd slab
                     const float* __restrict__ alpha_dev, float alpha_host,
                     const float* __restrict__ hl,   // [R, 2A]: mu | log_std
                     const float* __restrict__ prob,
@@ -718,6 +721,7 @@ void tg_bwd2_kernel(const float* __restrict__ dxc, int ld_dxc, int col0,
   float t = tanhf(p);
   float se = p - hl[(int64_t)b * 2 * A + a];   // std * eps
   float dpi = dxc[(int64_t)b * ld_dxc + col0 + a];
+  if (dxc2) dpi += dxc2[(int64_t)b * ld_dxc + col0 + a];
   float dl = alpha / B;
   float dp = dpi * act_limit * (1.f - t * t);
   float g_mu = dp + dl * t;
@@ -994,21 +998,23 @@ void mgemm(std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
            bool relu,
            std::vector<torch::Tensor> xs2, std::vector<torch::Tensor> ws2,
            std::vector<c10::optional<torch::Tensor>> masks2, int64_t K2,
-           int64_t x_off, int64_t x2_off) {
+           int64_t x_off, int64_t x2_off,
+           std::vector<int64_t> x_offs) {
   const int nz = (int)xs.size();
-  TORCH_CHECK(nz >= 1 && nz <= 2);
+  TORCH_CHECK(nz >= 1 && nz <= MAXZ);
   const bool sum2 = !xs2.empty();
   const bool has_mask = masks.size() && masks[0].has_value();
   MGemm g{};
   g.M = (int)M; g.N = (int)N; g.K = (int)K;
   g.lda = (int)lda; g.ldy = (int)ldy; g.K2 = (int)K2;
   for (int z = 0; z < nz; ++z) {
-    g.p[z].x = xs[z].data_ptr<float>() + x_off;
+    int64_t xo = x_offs.empty() ? x_off : x_offs[z];
+    g.p[z].x = xs[z].data_ptr<float>() + xo;
     g.p[z].w = ws[z].data_ptr<float>();
     g.p[z].bias = fptr(bs[z]);
     g.p[z].y = ys[z].data_ptr<float>();
     g.p[z].mask = masks.size() ? fptr(masks[z]) : nullptr;
-    if (g.p[z].mask) g.p[z].mask += x_off;
+    if (g.p[z].mask) g.p[z].mask += (x_offs.empty() ? x_off : x_offs[z]);
     if (sum2) {
       g.p[z].x2 = xs2[z].data_ptr<float>() + x2_off;
       g.p[z].w2 = ws2[z].data_ptr<float>();
@@ -1124,6 +1130,7 @@ void tg_fwd2(torch::Tensor hl, torch::Tensor out0,
 }
 
 void tg_bwd2(torch::Tensor dxc, int64_t col0,
+             c10::optional<torch::Tensor> dxc2,
              c10::optional<torch::Tensor> alpha_dev, double alpha_host,
              torch::Tensor hl, torch::Tensor prob,
              torch::Tensor dmu, torch::Tensor dls, int64_t B,
@@ -1131,6 +1138,7 @@ void tg_bwd2(torch::Tensor dxc, int64_t col0,
   const int A = (int)dmu.size(1);
   hipLaunchKernelGGL(tg_bwd2_kernel, dim3((int)B), dim3(64), 0, stream(),
                      dxc.data_ptr<float>(), (int)dxc.size(1), (int)col0,
+                     fptr(dxc2),
                      fptr(alpha_dev), (float)alpha_host,
                      hl.data_ptr<float>(),
                      prob.data_ptr<float>(), dmu.data_ptr<float>(),

@@ -82,7 +82,7 @@ class _NativeLinear(torch.autograd.Function):
         y = torch.empty(M, N, device=x.device, dtype=x.dtype)
         # pipelined multi-problem GEMM kernel (fused.hip), single problem
         ext.mgemm([x], [w], [b], [y], [None], M, N, K, K, N, relu,
-                  [], [], [], 0, 0, 0)
+                  [], [], [], 0, 0, 0, [])
         ctx.save_for_backward(x, w, y)
         ctx.relu = relu
         return y

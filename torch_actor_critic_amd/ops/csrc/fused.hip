@@ -397,12 +397,26 @@ struct WProb {
 struct WGemm {
   WProb p[2];
   int M, N, K, lddy, ldx;
+  int nz, m_chunk;           // split-M: blockIdx.z = slab * nz + z
+  float* part;               // [split][nz][N*K + N] partial slabs (or null)
 };
 
 template <bool BF16, bool MASK>
 __global__ __launch_bounds__(256)
 void mwgrad_kernel(WGemm g) {
-  const WProb& p = g.p[blockIdx.z];
+  const int zz = (int)blockIdx.z % g.nz;
+  const int slab = (int)blockIdx.z / g.nz;
+  const WProb& p = g.p[zz];
+  const int m_lo = slab * g.m_chunk;
+  const int m_hi = min(g.M, m_lo + g.m_chunk);
+  float* dw_out = p.dw;
+  float* db_out = p.db;
+  if (g.part) {
+    float* base = g.part
+        + ((int64_t)slab * g.nz + zz) * ((int64_t)g.N * g.K + g.N);
+    dw_out = base;
+    db_out = p.db ? base + (int64_t)g.N * g.K : nullptr;
+  }
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -418,7 +432,7 @@ void mwgrad_kernel(WGemm g) {
   if (tid < 64) dbs[tid] = 0.f;
   f32x4 acc[2][2] = {};
 
-  for (int i0 = 0; i0 < g.M; i0 += BK) {
+  for (int i0 = m_lo; i0 < m_hi; i0 += BK) {
     // A tile: as[n][i] = dYeff[i0+i][bn0+n]; staged transposed.
     // thread: i = tid&(BK-1)... BK may be 16 (fp32): use i = tid % BK.
     {
@@ -427,7 +441,7 @@ void mwgrad_kernel(WGemm g) {
         const int ic = tid & 63;
         const int nc0 = (tid >> 6) * 16;
         const int gi = i0 + ic;
-        bool interior = (i0 + 64 <= g.M) && (bn0 + 64 <= g.N)
+        bool interior = (i0 + 64 <= m_hi) && (bn0 + 64 <= g.N)
                         && ((g.lddy & 3) == 0) && ((bn0 & 3) == 0);
         float v[16];
         if (interior) {
@@ -453,7 +467,7 @@ void mwgrad_kernel(WGemm g) {
           for (int e = 0; e < 16; ++e) {
             int n = bn0 + nc0 + e;
             float val = 0.f;
-            if (gi < g.M && n < g.N) {
+            if (gi < m_hi && n < g.N) {
               val = p.dy[(int64_t)gi * g.lddy + n];
               if constexpr (MASK) {
                 val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
@@ -479,7 +493,7 @@ void mwgrad_kernel(WGemm g) {
           int n = bn0 + nc0 + e;
           int gi = i0 + ic;
           float val = 0.f;
-          if (gi < g.M && n < g.N) {
+          if (gi < m_hi && n < g.N) {
             val = p.dy[(int64_t)gi * g.lddy + n];
             if constexpr (MASK) {
               val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
@@ -496,7 +510,7 @@ void mwgrad_kernel(WGemm g) {
         const int ic = tid & 63;
         const int kc0 = (tid >> 6) * 16;
         const int gi = i0 + ic;
-        bool interior = (i0 + 64 <= g.M) && (bk0 + 64 <= g.K)
+        bool interior = (i0 + 64 <= m_hi) && (bk0 + 64 <= g.K)
                         && ((g.ldx & 3) == 0) && ((bk0 & 3) == 0);
         float v[16];
         if (interior) {
@@ -511,7 +525,7 @@ void mwgrad_kernel(WGemm g) {
           for (int e = 0; e < 16; ++e) {
             int k = bk0 + kc0 + e;
             int gi2 = i0 + ic;
-            v[e] = (gi2 < g.M && k < g.K)
+            v[e] = (gi2 < m_hi && k < g.K)
                        ? p.x[(int64_t)gi2 * g.ldx + k] : 0.f;
           }
         }
@@ -527,7 +541,7 @@ void mwgrad_kernel(WGemm g) {
           int k = bk0 + kc0 + e;
           int gi = i0 + ic;
           bs[(kc0 + e) * LDSF2 + ic] =
-              (gi < g.M && k < g.K) ? p.x[(int64_t)gi * g.ldx + k] : 0.f;
+              (gi < m_hi && k < g.K) ? p.x[(int64_t)gi * g.ldx + k] : 0.f;
         }
       }
     }
@@ -559,10 +573,30 @@ void mwgrad_kernel(WGemm g) {
         int gn = bn0 + wrow + mi * 16 + crow + r;
         int gk = bk0 + wcol + ni * 16 + ccol;
         if (gn < g.N && gk < g.K)
-          p.dw[(int64_t)gn * g.K + gk] = acc[mi][ni][r];
+          dw_out[(int64_t)gn * g.K + gk] = acc[mi][ni][r];
       }
-  if (p.db && blockIdx.y == 0 && tid < 64 && bn0 + tid < g.N)
-    p.db[bn0 + tid] = dbs[tid];
+  if (db_out && blockIdx.y == 0 && tid < 64 && bn0 + tid < g.N)
+    db_out[bn0 + tid] = dbs[tid];
+}
+
+// deterministic combine of split-M wgrad slabs: for each problem z,
+// dw[i] = sum_slab part[slab][z][i] (db appended after dw)
+__global__ __launch_bounds__(256)
+void mwgrad_combine_kernel(const float* __restrict__ part, WGemm g,
+                           int split) {
+  const int64_t per = (int64_t)g.N * g.K + g.N;
+  const int64_t total = per * g.nz;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int z = (int)(i / per);
+    const int64_t off = i % per;
+    float sm = 0.f;
+    for (int sl = 0; sl < split; ++sl)
+      sm += part[((int64_t)sl * g.nz + z) * per + off];
+    if (off < (int64_t)g.N * g.K) g.p[z].dw[off] = sm;
+    else if (g.p[z].db) g.p[z].db[off - (int64_t)g.N * g.K] = sm;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1060,8 +1094,23 @@ void mwgrad(std::vector<torch::Tensor> dys,
     g.p[z].dw = dws[z].data_ptr<float>();
     g.p[z].db = dbs[z].numel() ? dbs[z].data_ptr<float>() : nullptr;
   }
-  dim3 grid((N + TB - 1) / TB, (K + TB - 1) / TB, nz);
   const bool bf16 = *g_bf16_flag;
+  const int BKc = bf16 ? BKB2 : BKF2;
+  const int chunks = (int)((M + BKc - 1) / BKc);
+  const int tiles = (int)(((N + TB - 1) / TB) * ((K + TB - 1) / TB) * nz);
+  int split = std::max(1, std::min(chunks, (384 + tiles - 1) / tiles));
+  int m_chunk = (int)(((M + split - 1) / split + BKc - 1) / BKc * BKc);
+  split = (int)((M + m_chunk - 1) / m_chunk);
+  g.nz = nz;
+  g.m_chunk = m_chunk;
+  g.part = nullptr;
+  torch::Tensor part;
+  if (split > 1) {
+    part = torch::empty({(int64_t)split * nz, (int64_t)N * K + N},
+                        dws[0].options());
+    g.part = part.data_ptr<float>();
+  }
+  dim3 grid((N + TB - 1) / TB, (K + TB - 1) / TB, nz * split);
   if (bf16) {
     if (has_mask)
       hipLaunchKernelGGL((mwgrad_kernel<true, true>), grid, dim3(256), 0,
@@ -1076,6 +1125,12 @@ void mwgrad(std::vector<torch::Tensor> dys,
     else
       hipLaunchKernelGGL((mwgrad_kernel<false, false>), grid, dim3(256), 0,
                          stream(), g);
+  }
+  if (split > 1) {
+    int64_t total = ((int64_t)N * K + N) * nz;
+    int blocks = (int)std::min<int64_t>((total + 255) / 256, 512);
+    hipLaunchKernelGGL(mwgrad_combine_kernel, dim3(blocks), dim3(256), 0,
+                       stream(), part.data_ptr<float>(), g, split);
   }
 }
 

@@ -341,3 +341,32 @@ def test_main_cli_on_gpu(tmp_path):
     assert len(runs) == 1
     art = tmp_path / "mlruns" / "0" / runs[0] / "artifacts"
     assert (art / "actor" / "data" / "model.pth").exists()
+
+
+def test_engine_split_graph_structure(monkeypatch):
+    """The data-parallel 3-graph capture (used at world>1, where the two
+    flat-bucket all-reduces run between segments) must behave exactly
+    like the single-graph capture — forced at world=1 via env var."""
+    monkeypatch.setenv("TAC_AMD_SPLIT_GRAPHS", "1")
+    sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+        _setup()
+    rng = np.random.default_rng(8)
+    buf.store_batch(rng.standard_normal((512, O)).astype(np.float32),
+                    rng.standard_normal((512, A)).astype(np.float32),
+                    rng.standard_normal(512).astype(np.float32),
+                    rng.standard_normal((512, O)).astype(np.float32),
+                    np.zeros(512, dtype=np.float32))
+    from torch_actor_critic_amd.algo.engine import FusedSACEngine
+    eng2 = FusedSACEngine(sac, actor, critic, target, buf, pi_opt, q_opt,
+                          target_flat, B, torch.device(DEV), sample=True,
+                          capture=True, philox_seed=17)
+    assert eng2._graphs is not None and eng2.graph is None
+    p0 = pi_opt.fp.flat.clone()
+    for _ in range(5):
+        eng2.step()
+    torch.cuda.synchronize()
+    assert not torch.allclose(p0, pi_opt.fp.flat)
+    assert torch.isfinite(pi_opt.fp.flat).all()
+    assert torch.isfinite(q_opt.fp.flat).all()
+    lq, lp = eng2.read_and_reset_losses(5)
+    assert np.isfinite(lq) and np.isfinite(lp) and lq > 0

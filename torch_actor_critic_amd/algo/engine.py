@@ -450,6 +450,7 @@ class FusedSACEngine:
         self._phase_finish()
 
     def _capture(self):
+        import os
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -457,7 +458,12 @@ class FusedSACEngine:
                 self._run_once()
         torch.cuda.current_stream().wait_stream(s)
 
-        if self.world == 1:
+        # TAC_AMD_SPLIT_GRAPHS=1 forces the data-parallel 3-graph
+        # structure at world=1 so the multi-GPU capture path is testable
+        # on a single GPU
+        split = (self.world > 1
+                 or os.environ.get("TAC_AMD_SPLIT_GRAPHS") == "1")
+        if not split:
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
                 self._phase_critic()

@@ -70,7 +70,10 @@ class GraphedSACUpdate:
         torch.cuda.current_stream().wait_stream(s)
 
         # -- capture ----------------------------------------------------
-        if self.world == 1:
+        import os
+        split = (self.world > 1
+                 or os.environ.get("TAC_AMD_SPLIT_GRAPHS") == "1")
+        if not split:
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
                 self._phase_critic()

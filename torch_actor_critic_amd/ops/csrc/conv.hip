@@ -470,7 +470,8 @@ std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
   // split the batch reduction so ~256+ blocks are in flight
   const int BKc = 64;
   int max_split = (M + BKc - 1) / BKc;
-  int split = std::max(1, std::min(max_split, (512 + tiles - 1) / tiles));
+  int split = std::max(1, std::min({max_split,
+                                    (256 + tiles - 1) / tiles, 64}));
   int m_chunk = ((M + split - 1) / split + BKc - 1) / BKc * BKc;
   split = (M + m_chunk - 1) / m_chunk;
   auto part = torch::empty({split, (int64_t)d.OC * K + d.OC}, w.options());

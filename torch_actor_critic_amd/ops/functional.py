@@ -91,8 +91,21 @@ class _NativeLinear(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
         ext = require_extension()
-        dx, dw, db = ext.linear_bwd(dy.contiguous(), x, w, y, ctx.relu,
-                                    ctx.needs_input_grad[0])
+        dy = dy.contiguous()
+        M, K = x.shape
+        N = w.shape[0]
+        ymask = y if ctx.relu else None
+        dw = torch.empty_like(w)
+        db = torch.empty(N, device=w.device, dtype=w.dtype)
+        # pipelined coalesced wgrad (split-M at large batch) + fused db
+        ext.mwgrad([dy], [ymask], [x], [dw], [db], M, N, K, N, K, 0)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            wt = torch.empty(K, N, device=w.device, dtype=w.dtype)
+            ext.transpose_multi([w], [wt])
+            dx = torch.empty_like(x)
+            ext.mgemm([dy], [wt], [None], [dx], [ymask], M, K, N, N, K,
+                      False, [], [], [], 0, 0, 0, [])
         return dx, dw, db, None
 
 

@@ -151,23 +151,48 @@ void conv_fwd_kernel(const float* __restrict__ x, const float* __restrict__ w,
   const int c00 = (tid >> 6) * EL;
   const int m_my = bm0 + row;
   const MDec md = mdec(m_my < M ? m_my : 0, d);
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  const int n_my = bn0 + row;
+  const float* wrow_p = n_my < d.OC ? w + (int64_t)n_my * K : nullptr;
+
+  // T14 register-staged pipeline: gather tile t into regs, write LDS,
+  // prefetch tile t+1 while the MFMAs run.
+  float va[EL], vb[EL];
+  auto load_chunk = [&](int k0) {
     KDec kd = kdec((k0 + c00) < K ? (k0 + c00) : 0, d);
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
-      int c = c00 + e;
-      int k = k0 + c;
-      float va = 0.f;
+      int k = k0 + c00 + e;
+      float v = 0.f;
       if (m_my < M && k < K) {
         int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
-        va = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
+        v = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
       }
-      lds_put<BF16>(xs, row, c, va);
-      int n = bn0 + row;
-      float vb = (n < d.OC && k < K) ? w[(int64_t)n * K + k] : 0.f;
-      lds_put<BF16>(ws, row, c, vb);
+      va[e] = v;
       kinc(kd, d);
     }
+    if (wrow_p && ((K & 3) == 0) && k0 + c00 + EL <= K) {
+      const float4* src = (const float4*)(wrow_p + k0 + c00);
+#pragma unroll
+      for (int q = 0; q < EL / 4; ++q) {
+        float4 f = src[q];
+        vb[q*4+0]=f.x; vb[q*4+1]=f.y; vb[q*4+2]=f.z; vb[q*4+3]=f.w;
+      }
+    } else {
+#pragma unroll
+      for (int e = 0; e < EL; ++e) {
+        int k = k0 + c00 + e;
+        vb[e] = (wrow_p && k < K) ? wrow_p[k] : 0.f;
+      }
+    }
+  };
+  load_chunk(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      lds_put<BF16>(xs, row, c00 + e, va[e]);
+      lds_put<BF16>(ws, row, c00 + e, vb[e]);
+    }
+    if (k0 + BK < K) load_chunk(k0 + BK);
     __syncthreads();
     mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
     __syncthreads();

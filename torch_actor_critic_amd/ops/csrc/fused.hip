@@ -1145,7 +1145,11 @@ void transpose_multi(std::vector<torch::Tensor> ws,
     t.N[i] = (int)ws[i].size(0);
     t.K[i] = (int)ws[i].size(1);
   }
-  t.blocks_per_layer = 8;
+  int64_t max_elems = 0;
+  for (int i = 0; i < t.n_layers; ++i)
+    max_elems = std::max(max_elems, (int64_t)t.N[i] * t.K[i]);
+  t.blocks_per_layer = (int)std::min<int64_t>(
+      128, std::max<int64_t>(8, (max_elems + 4095) / 4096));
   hipLaunchKernelGGL(transpose_multi_kernel,
                      dim3(t.n_layers * t.blocks_per_layer), dim3(256), 0,
                      stream(), t);

@@ -256,35 +256,57 @@ void conv_dgrad_kernel(const float* __restrict__ dy,
     iy0 = t1 % d.IH;
     b0 = t1 / d.IH;
   }
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // k = (oc, ky, kx) walked incrementally (KW/KH in place of d's
-    // meanings: here k indexes (oc, ky, kx))
+  const int n_my = bn0 + row;
+  const float* wtrow = n_my < d.IC ? wt + (int64_t)n_my * K : nullptr;
+
+  // T14 register-staged pipeline (prefetch chunk t+1 during MFMAs)
+  float va[EL], vb[EL];
+  auto load_chunk = [&](int k0) {
     int kx = (k0 + c00) % d.KW;
     int t2 = (k0 + c00) / d.KW;
     int ky = t2 % d.KH;
     int oc = t2 / d.KH;
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
-      int c = c00 + e;
-      int k = k0 + c;
-      float va = 0.f;
+      int k = k0 + c00 + e;
+      float v = 0.f;
       if (m_my < M && k < K) {
         const int ry = iy0 - ky, rx = ix0 - kx;
         if (ry >= 0 && rx >= 0 && ry % d.S == 0 && rx % d.S == 0) {
           const int oy = ry / d.S, ox = rx / d.S;
           if (oy < d.OH && ox < d.OW) {
             int64_t idx = (((int64_t)b0 * d.OC + oc) * d.OH + oy) * d.OW + ox;
-            va = dy[idx];
-            if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
+            v = dy[idx];
+            if constexpr (MASK) v = ymask[idx] > 0.f ? v : 0.f;
           }
         }
       }
-      lds_put<BF16>(xs, row, c, va);
-      int n = bn0 + row;
-      float vb = (n < d.IC && k < K) ? wt[(int64_t)n * K + k] : 0.f;
-      lds_put<BF16>(ws, row, c, vb);
+      va[e] = v;
       if (++kx == d.KW) { kx = 0; if (++ky == d.KH) { ky = 0; ++oc; } }
     }
+    if (wtrow && ((K & 3) == 0) && k0 + c00 + EL <= K) {
+      const float4* src = (const float4*)(wtrow + k0 + c00);
+#pragma unroll
+      for (int q = 0; q < EL / 4; ++q) {
+        float4 f = src[q];
+        vb[q*4+0]=f.x; vb[q*4+1]=f.y; vb[q*4+2]=f.z; vb[q*4+3]=f.w;
+      }
+    } else {
+#pragma unroll
+      for (int e = 0; e < EL; ++e) {
+        int k = k0 + c00 + e;
+        vb[e] = (wtrow && k < K) ? wtrow[k] : 0.f;
+      }
+    }
+  };
+  load_chunk(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      lds_put<BF16>(xs, row, c00 + e, va[e]);
+      lds_put<BF16>(ws, row, c00 + e, vb[e]);
+    }
+    if (k0 + BK < K) load_chunk(k0 + BK);
     __syncthreads();
     mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
     __syncthreads();

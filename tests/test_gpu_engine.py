@@ -370,3 +370,87 @@ def test_engine_split_graph_structure(monkeypatch):
     assert torch.isfinite(q_opt.fp.flat).all()
     lq, lp = eng2.read_and_reset_losses(5)
     assert np.isfinite(lq) and np.isfinite(lp) and lq > 0
+
+
+def test_engine_parity_humanoid_dims():
+    """Full-update parity at Humanoid-v4 shapes (obs 376, act 17 —
+    odd 393-wide concat exercises the scalar staging edges)."""
+    global B, O, A, HID
+    oldB, oldO, oldA, oldH = B, O, A, HID
+    B, O, A, HID = 32, 376, 17, [256, 256]
+    try:
+        sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+            _setup(batch=B)
+        from torch_actor_critic_amd.ops import require_extension
+        ext = require_extension()
+        actor_cpu = copy.deepcopy(actor).cpu()
+        critic_cpu = copy.deepcopy(critic).cpu()
+        target_cpu = copy.deepcopy(target).cpu()
+        torch.manual_seed(6)
+        s = torch.randn(B, O)
+        a = torch.rand(B, A) * 2 - 1
+        r = torch.randn(B)
+        ns = torch.randn(B, O)
+        d = torch.zeros(B)
+        eng.load_batch(s.to(DEV), a.to(DEV), r.to(DEV), ns.to(DEV),
+                       d.to(DEV))
+        eng._run_once()
+        torch.cuda.synchronize()
+        eps = ext.tg_eps(1, 0, 2 * B, A, eng.prob).cpu()
+        ref = _eager_reference(actor_cpu, critic_cpu, target_cpu,
+                               s, a, r, ns, d, eps)
+        assert abs(float(eng.loss_q_acc.item()) - ref["loss_q"]) \
+            < 5e-3 * max(1, abs(ref["loss_q"]))
+        cg = q_opt.fp.flat_grad.cpu()
+        assert torch.allclose(cg, ref["cgrad"], atol=1e-4, rtol=1e-3), \
+            (cg - ref["cgrad"]).abs().max()
+        ag = pi_opt.fp.flat_grad.cpu()
+        assert torch.allclose(ag, ref["agrad"], atol=1e-4, rtol=1e-3), \
+            (ag - ref["agrad"]).abs().max()
+    finally:
+        B, O, A, HID = oldB, oldO, oldA, oldH
+
+
+def test_sac_train_bf16_learns_pendulum():
+    """End-to-end learning with the bf16 MFMA compute mode (the bench
+    dtype): SAC must still solve Pendulum."""
+    from torch_actor_critic_amd import envs
+    from torch_actor_critic_amd.algo.sac import SAC
+    from torch_actor_critic_amd.buffer.replay import ReplayBuffer
+    from torch_actor_critic_amd.models.mlp import Actor, DoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd.ops import functional as Fo
+
+    torch.manual_seed(0)
+    np.random.seed(0)
+    Fo.set_philox_seed(0)
+    Fo.set_compute_dtype("bf16")
+    try:
+        device = torch.device(DEV)
+        env = envs.make("Pendulum-v1")
+        env.seed(0)
+        actor = Actor(3, 1, [64, 64], act_limit=2.0).to(device)
+        critic = DoubleCritic(3, 1, [64, 64]).to(device)
+        buf = ReplayBuffer(20000, 3, 1, device=device)
+        pi_opt = FlatAdam(actor, lr=1e-3)
+        q_opt = FlatAdam(critic, lr=1e-3)
+        sac = SAC(alpha=0.1, gamma=0.99, polyak=0.995, reward_scale=1.0,
+                  epochs=1, batch_size=64, start_steps=500,
+                  steps_per_epoch=6000, max_ep_len=200, update_after=500,
+                  update_every=50, save_every=1000)
+        sac.train(0, env, actor, critic, buf, pi_opt, q_opt,
+                  render=False, logging=False)
+        rets = []
+        for _ in range(5):
+            state = env.reset()
+            ep, done = 0.0, False
+            while not done:
+                with torch.no_grad():
+                    act, _ = actor(torch.as_tensor(state, device=device),
+                                   deterministic=True, with_logprob=False)
+                state, rr, done, _ = env.step(act.cpu().numpy())
+                ep += rr
+            rets.append(ep)
+    finally:
+        Fo.set_compute_dtype("fp32")
+    assert float(np.mean(rets)) > -700.0, rets

@@ -178,6 +178,16 @@ class FusedSACEngine:
             self.alpha_dev = None
             self.mean_logp = None
 
+        # side stream for wgrads (no consumer until Adam): they overlap
+        # the dgrad chain; fork/join via events (captured as graph deps).
+        # TAC_AMD_WGRAD_STREAM=0 disables for A/B.
+        self._use_side = _os.environ.get("TAC_AMD_WGRAD_STREAM") != "0"
+        self._s2 = torch.cuda.Stream()
+        self._fork_evs = [torch.cuda.Event() for _ in range(12)]
+        self._join_evs = [torch.cuda.Event() for _ in range(4)]
+        self._ev_i = 0
+        self._join_i = 0
+
         torch.cuda.synchronize()
 
         # ---- capture ---------------------------------------------------
@@ -189,6 +199,29 @@ class FusedSACEngine:
     # ------------------------------------------------------------------
     # recorded phases
     # ------------------------------------------------------------------
+
+    def _side(self, fn):
+        """Run fn's kernels on the side stream, ordered after the work
+        already enqueued on the main stream."""
+        if not self._use_side:
+            fn()
+            return
+        ev = self._fork_evs[self._ev_i % len(self._fork_evs)]
+        self._ev_i += 1
+        ev.record()
+        with torch.cuda.stream(self._s2):
+            self._s2.wait_event(ev)
+            fn()
+
+    def _join_side(self):
+        """Main stream waits for everything enqueued on the side stream."""
+        if not self._use_side:
+            return
+        ev = self._join_evs[self._join_i % len(self._join_evs)]
+        self._join_i += 1
+        with torch.cuda.stream(self._s2):
+            ev.record()
+        torch.cuda.current_stream().wait_event(ev)
 
     def _mg(self, xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
             xs2=None, ws2=None, masks2=None, K2=0, x_off=0, x2_off=0,
@@ -332,11 +365,15 @@ class FusedSACEngine:
             else:
                 x_in = [self.XC, self.XC]
                 ldx, xoff = OC, 0
-            ext.mwgrad(d, masks, x_in,
-                       [self.cw[z][i][0].grad for z in range(2)],
-                       [self.cw[z][i][1].grad for z in range(2)],
-                       B, self.c_w[i], (self.c_w[i - 1] if i > 0 else OC),
-                       self.c_w[i], ldx, xoff)
+            dd, mm, xx = list(d), list(masks), list(x_in)
+            ii = i
+            self._side(lambda: ext.mwgrad(
+                dd, mm, xx,
+                [self.cw[z][ii][0].grad for z in range(2)],
+                [self.cw[z][ii][1].grad for z in range(2)],
+                B, self.c_w[ii],
+                (self.c_w[ii - 1] if ii > 0 else OC),
+                self.c_w[ii], ldx, xoff))
             if i > 0:
                 if i == nL - 1 and fuse:
                     pass  # dy2 already produced by the fused loss kernel
@@ -347,6 +384,8 @@ class FusedSACEngine:
                              masks, B, self.c_w[i - 1], self.c_w[i],
                              self.c_w[i], self.c_w[i - 1], False)
                 d = [self.dc[z][i - 1] for z in range(2)]
+        # critic grads must be complete before all-reduce / Adam
+        self._join_side()
 
     def _phase_policy(self):
         ext = self.ext
@@ -398,9 +437,10 @@ class FusedSACEngine:
         h_last = self.a_hidden[-1]
         a_last = self.a_act[-1]
         (wm, bm), (wl, bl) = self.head_w
-        ext.mwgrad([self.dmu, self.dls], [None, None], [a_last, a_last],
-                   [wm.grad, wl.grad], [bm.grad, bl.grad],
-                   B, A, h_last, A, h_last, 0)
+        self._side(lambda: ext.mwgrad(
+            [self.dmu, self.dls], [None, None], [a_last, a_last],
+            [wm.grad, wl.grad], [bm.grad, bl.grad],
+            B, A, h_last, A, h_last, 0))
         self._mg([self.dmu], [self.hwt[0]], [None], [self.da[-1]], [None],
                  B, h_last, A, A, h_last, False,
                  xs2=[self.dls], ws2=[self.hwt[1]], masks2=[None], K2=A)
@@ -415,14 +455,17 @@ class FusedSACEngine:
                 x_in, ldx, xoff = self.XC, OC, 0
                 kin = O
             (w, b) = self.aw[i]
-            ext.mwgrad([d], [mask], [x_in], [w.grad], [b.grad],
-                       B, self.a_hidden[i], kin, self.a_hidden[i], ldx,
-                       xoff)
+            dd, mm2, xx2, ii = d, mask, x_in, i
+            self._side(lambda: ext.mwgrad(
+                [dd], [mm2], [xx2], [w.grad], [b.grad],
+                B, self.a_hidden[ii], kin, self.a_hidden[ii], ldx, xoff))
             if i > 0:
                 self._mg([d], [self.awt[i]], [None], [self.da[i - 1]],
                          [mask], B, self.a_hidden[i - 1], self.a_hidden[i],
                          self.a_hidden[i], self.a_hidden[i - 1], False)
                 d = self.da[i - 1]
+        # actor grads must be complete before all-reduce / Adam
+        self._join_side()
 
     def _phase_finish(self):
         ext = self.ext

@@ -178,10 +178,12 @@ class FusedSACEngine:
             self.alpha_dev = None
             self.mean_logp = None
 
-        # side stream for wgrads (no consumer until Adam): they overlap
-        # the dgrad chain; fork/join via events (captured as graph deps).
-        # TAC_AMD_WGRAD_STREAM=0 disables for A/B.
-        self._use_side = _os.environ.get("TAC_AMD_WGRAD_STREAM") != "0"
+        # Side-stream wgrads (overlapping the dgrad chain via captured
+        # fork/join events) measured MUCH slower in-graph: A/B 3732 vs
+        # 5464 updates/s — the cross-queue dependency overhead dwarfs the
+        # ~5 us kernels at this scale.  Kept for study via
+        # TAC_AMD_WGRAD_STREAM=1; default off.
+        self._use_side = _os.environ.get("TAC_AMD_WGRAD_STREAM") == "1"
         self._s2 = torch.cuda.Stream()
         self._fork_evs = [torch.cuda.Event() for _ in range(12)]
         self._join_evs = [torch.cuda.Event() for _ in range(4)]

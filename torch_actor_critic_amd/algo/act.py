@@ -47,12 +47,36 @@ class ActKernel:
         self.obs_pin = torch.zeros(obs_dim, pin_memory=True)
         self.act_out = torch.zeros(act_dim, device=device)
         self.act_pin = torch.zeros(act_dim, pin_memory=True)
+        self.flag_pin = torch.zeros(1, dtype=torch.int32, pin_memory=True)
+        self.flag_np = self.flag_pin.numpy()
         self.ev = torch.cuda.Event()
+        import os
+        self._pinned_ok = os.environ.get("TAC_AMD_ACT_PINNED", "1") != "0"''
 
     def act(self, state: np.ndarray) -> np.ndarray:
         self.obs_pin.copy_(torch.from_numpy(np.asarray(state,
                                                        dtype=np.float32)))
         self.obs_in.copy_(self.obs_pin, non_blocking=True)
+        if self._pinned_ok:
+            # the kernel writes the action AND a completion flag straight
+            # into host-pinned memory (system-scope release): the host
+            # spins on the flag — no D2H copy, no event sync
+            self.flag_pin[0] = 0
+            self.ext.act_step_pinned(
+                self.obs_in, self.ws, self.bs, self.wmu, self.bmu,
+                self.wls, self.bls, self.act_out, self.act_pin,
+                self.flag_pin, self.ctr, self.seed, self.act_limit,
+                self.lo, self.hi)
+            fl = self.flag_np
+            for _ in range(2_000_000):
+                if fl[0] != 0:
+                    return self.act_pin.numpy().copy()
+            # flag never landed: drain the stream once, then give up on
+            # the pinned path for this session
+            torch.cuda.synchronize()
+            if fl[0] != 0:
+                return self.act_pin.numpy().copy()
+            self._pinned_ok = False
         self.ext.act_step(self.obs_in, self.ws, self.bs, self.wmu,
                           self.bmu, self.wls, self.bls, self.act_out,
                           self.ctr, self.seed, self.act_limit, self.lo,

@@ -896,8 +896,16 @@ struct ActArgs {
   float act_limit, lo, hi;
 };
 
+// completion flag written to host-visible pinned memory after the
+// action stores (system-scope release), so the host can spin instead of
+// paying a D2H copy + event synchronization per env step.
+struct ActPinned {
+  float* out_host;        // pinned [A]
+  int* flag_host;         // pinned [1]
+};
+
 __global__ __launch_bounds__(256)
-void act_kernel(ActArgs a) {
+void act_kernel(ActArgs a, ActPinned hp) {
   __shared__ __attribute__((aligned(16))) float buf[2][ACT_MAXW];
   __shared__ unsigned long long ctr_s;
   const int tid = threadIdx.x;
@@ -941,7 +949,17 @@ void act_kernel(ActArgs a) {
     float u0 = (r.x + 1.f) * 2.3283064365386963e-10f;
     float u1 = (r.y + 1.f) * 2.3283064365386963e-10f;
     float eps = sqrtf(-2.f * logf(u0)) * __cosf(6.283185307179586f * u1);
-    a.action[tid] = tanhf(mu + expf(ls) * eps) * a.act_limit;
+    float act = tanhf(mu + expf(ls) * eps) * a.act_limit;
+    a.action[tid] = act;
+    if (hp.out_host) hp.out_host[tid] = act;
+  }
+  if (hp.flag_host) {
+    __threadfence_system();
+    __syncthreads();
+    if (tid == 0) {
+      __threadfence_system();
+      *(volatile int*)hp.flag_host = 1;
+    }
   }
 }
 
@@ -1312,7 +1330,49 @@ void act_step(torch::Tensor x, std::vector<torch::Tensor> ws,
   a.act_limit = (float)act_limit;
   a.lo = (float)lo;
   a.hi = (float)hi;
-  hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(), a);
+  hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(), a,
+                     ActPinned{nullptr, nullptr});
+}
+
+// act_step writing the action AND a completion flag to host-pinned
+// memory (host spins on the flag; no D2H copy / event sync per step)
+void act_step_pinned(torch::Tensor x, std::vector<torch::Tensor> ws,
+                     std::vector<torch::Tensor> bs, torch::Tensor wmu,
+                     torch::Tensor bmu, torch::Tensor wls,
+                     torch::Tensor bls, torch::Tensor action,
+                     torch::Tensor out_host, torch::Tensor flag_host,
+                     torch::Tensor ctr, int64_t seed, double act_limit,
+                     double lo, double hi) {
+  TORCH_CHECK(out_host.is_pinned() && flag_host.is_pinned(),
+              "host buffers must be pinned");
+  ActArgs a{};
+  a.n_layers = (int)ws.size();
+  TORCH_CHECK(a.n_layers <= ACT_MAXL);
+  a.x = x.data_ptr<float>();
+  a.O = (int)x.numel();
+  a.A = (int)action.numel();
+  int maxw = a.O;
+  for (int L = 0; L < a.n_layers; ++L) {
+    a.w[L] = ws[L].data_ptr<float>();
+    a.b[L] = bs[L].data_ptr<float>();
+    a.width[L] = (int)ws[L].size(0);
+    TORCH_CHECK(a.width[L] <= 256);
+    maxw = std::max(maxw, a.width[L]);
+  }
+  TORCH_CHECK(maxw <= ACT_MAXW && a.A <= 256);
+  a.wmu = wmu.data_ptr<float>();
+  a.bmu = bmu.data_ptr<float>();
+  a.wls = wls.data_ptr<float>();
+  a.bls = bls.data_ptr<float>();
+  a.action = action.data_ptr<float>();
+  a.ctr = ctr.data_ptr<int64_t>();
+  a.seed = (uint64_t)seed;
+  a.act_limit = (float)act_limit;
+  a.lo = (float)lo;
+  a.hi = (float)hi;
+  hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(), a,
+                     ActPinned{out_host.data_ptr<float>(),
+                               flag_host.data_ptr<int>()});
 }
 
 void bump3(torch::Tensor a, c10::optional<torch::Tensor> b,
@@ -1350,4 +1410,5 @@ void register_fused(pybind11::module_& m) {
   m.def("adam_t", &fused::adam_t);
   m.def("act_step", &fused::act_step);
   m.def("bump3", &fused::bump3);
+  m.def("act_step_pinned", &fused::act_step_pinned);
 }

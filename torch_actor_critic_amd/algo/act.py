@@ -56,14 +56,14 @@ class ActKernel:
     def act(self, state: np.ndarray) -> np.ndarray:
         self.obs_pin.copy_(torch.from_numpy(np.asarray(state,
                                                        dtype=np.float32)))
-        self.obs_in.copy_(self.obs_pin, non_blocking=True)
         if self._pinned_ok:
-            # the kernel writes the action AND a completion flag straight
-            # into host-pinned memory (system-scope release): the host
-            # spins on the flag — no D2H copy, no event sync
+            # zero-copy I/O: the kernel READS the observation from and
+            # WRITES the action + a system-scope completion flag to
+            # host-pinned memory; the host spins on the flag — no H2D/D2H
+            # copies, no event sync
             self.flag_pin[0] = 0
             self.ext.act_step_pinned(
-                self.obs_in, self.ws, self.bs, self.wmu, self.bmu,
+                self.obs_pin, self.ws, self.bs, self.wmu, self.bmu,
                 self.wls, self.bls, self.act_out, self.act_pin,
                 self.flag_pin, self.ctr, self.seed, self.act_limit,
                 self.lo, self.hi)
@@ -77,6 +77,7 @@ class ActKernel:
             if fl[0] != 0:
                 return self.act_pin.numpy().copy()
             self._pinned_ok = False
+        self.obs_in.copy_(self.obs_pin, non_blocking=True)
         self.ext.act_step(self.obs_in, self.ws, self.bs, self.wmu,
                           self.bmu, self.wls, self.bls, self.act_out,
                           self.ctr, self.seed, self.act_limit, self.lo,

@@ -218,7 +218,11 @@ class SAC:
         from ..models.mlp import DoubleCritic as MlpDoubleCritic
         if (type(actor) is MlpActor and type(critic) is MlpDoubleCritic
                 and isinstance(buffer, ReplayBuffer)
-                and buffer.act_dim <= 64):
+                and buffer.act_dim <= 64
+                # the engine implements the standard (state-sampled)
+                # policy loss; the reference's next-state quirk (Q2)
+                # runs on the autograd graph below, which honors it
+                and not self.reference_pi_loss):
             try:
                 from .engine import FusedSACEngine
                 self._graph = FusedSACEngine(

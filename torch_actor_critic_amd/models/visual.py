@@ -52,11 +52,17 @@ class Linear(nn.Linear):
 
 class Conv2d(nn.Conv2d):
     """nn.Conv2d whose forward dispatches to the hand-written gfx950
-    implicit-GEMM MFMA kernels on GPU (ops/csrc/conv.hip); state-dict
-    identical to nn.Conv2d."""
+    implicit-GEMM MFMA kernels on GPU (ops/csrc/conv.hip), optionally
+    with the trailing ReLU fused in; state-dict identical to nn.Conv2d
+    (ReLU has no parameters)."""
+
+    def __init__(self, *args, fuse_relu: bool = False, **kw):
+        super().__init__(*args, **kw)
+        self.fuse_relu = fuse_relu
 
     def forward(self, x):
-        return Fo.conv2d(x, self.weight, self.bias, self.stride[0])
+        return Fo.conv2d(x, self.weight, self.bias, self.stride[0],
+                         relu=self.fuse_relu)
 
 
 def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
@@ -67,11 +73,15 @@ def simple_cnn(input_shape, filters=[32, 64, 64], kernel_sizes=[8, 4, 3],
     channels = input_shape[0]
     model = nn.Sequential()
     sizes = [channels] + list(filters)
+    fuse = activation is nn.ReLU
     for i in range(len(sizes) - 1):
         model.add_module(f"conv_{i}",
                          Conv2d(sizes[i], sizes[i + 1], kernel_sizes[i],
-                                strides[i]))
-        model.add_module(f"relu_{i}", activation())
+                                strides[i], fuse_relu=fuse))
+        # ReLU fuses into the conv kernel; keep the module slot (no
+        # params) so the Sequential structure matches the reference
+        model.add_module(f"relu_{i}",
+                         nn.Identity() if fuse else activation())
     flat = calculate_size(input_shape, filters, kernel_sizes, strides)
     model.add_module("flatten", nn.Flatten())
     model.add_module("linear", Linear(flat, dense_size))

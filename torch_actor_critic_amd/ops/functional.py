@@ -134,35 +134,46 @@ def mlp_forward(x, layers, relu_last: bool = True):
 
 class _NativeConv2d(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, b, stride):
+    def forward(ctx, x, w, b, stride, relu):
         ext = require_extension()
         y = ext.conv2d_fwd(x.contiguous(), w.contiguous(),
                            b.contiguous() if b is not None else None,
-                           stride, False)
-        ctx.save_for_backward(x, w)
+                           stride, relu)
+        if relu:
+            ctx.save_for_backward(x, w, y)
+        else:
+            ctx.save_for_backward(x, w)
         ctx.stride = stride
+        ctx.relu = relu
         ctx.has_bias = b is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w = ctx.saved_tensors
+        if ctx.relu:
+            x, w, y = ctx.saved_tensors
+            ymask = y
+        else:
+            x, w = ctx.saved_tensors
+            ymask = None
         ext = require_extension()
         dy = dy.contiguous()
         dx = None
         if ctx.needs_input_grad[0]:
             oc, ic, kh, kw = w.shape
             wt = w.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw).contiguous()
-            dx = ext.conv2d_dgrad(dy, None, wt, x, w, ctx.stride)
-        dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
-        return dx, dw, (db if ctx.has_bias else None), None
+            dx = ext.conv2d_dgrad(dy, ymask, wt, x, w, ctx.stride)
+        dw, db = ext.conv2d_wgrad(dy, ymask, x, w, ctx.stride)
+        return dx, dw, (db if ctx.has_bias else None), None, None
 
 
-def conv2d(x, w, b, stride: int):
-    """Valid-padding conv with square stride (the reference CNN family)."""
+def conv2d(x, w, b, stride: int, relu: bool = False):
+    """Valid-padding conv with square stride (the reference CNN family),
+    optionally with the trailing ReLU fused into the kernel."""
     if use_native(x, w):
-        return _NativeConv2d.apply(x, w, b, int(stride))
-    return F.conv2d(x, w, b, stride=stride)
+        return _NativeConv2d.apply(x, w, b, int(stride), relu)
+    y = F.conv2d(x, w, b, stride=stride)
+    return F.relu(y) if relu else y
 
 
 # ---------------------------------------------------------------------------

@@ -150,20 +150,21 @@ class _NativeConv2d(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        if ctx.relu:
-            x, w, y = ctx.saved_tensors
-            ymask = y
-        else:
-            x, w = ctx.saved_tensors
-            ymask = None
         ext = require_extension()
         dy = dy.contiguous()
+        if ctx.relu:
+            # one elementwise relu-backward beats masked gathers inside
+            # the conv kernels (measured)
+            x, w, y = ctx.saved_tensors
+            dy = torch.ops.aten.threshold_backward(dy, y, 0)
+        else:
+            x, w = ctx.saved_tensors
         dx = None
         if ctx.needs_input_grad[0]:
             oc, ic, kh, kw = w.shape
             wt = w.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw).contiguous()
-            dx = ext.conv2d_dgrad(dy, ymask, wt, x, w, ctx.stride)
-        dw, db = ext.conv2d_wgrad(dy, ymask, x, w, ctx.stride)
+            dx = ext.conv2d_dgrad(dy, None, wt, x, w, ctx.stride)
+        dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
         return dx, dw, (db if ctx.has_bias else None), None, None
 
 

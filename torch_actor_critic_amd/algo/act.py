@@ -26,7 +26,7 @@ class ActKernel:
     def __init__(self, actor, obs_dim: int, act_dim: int,
                  device: torch.device, philox_seed: int = 0):
         from ..models.mlp import Actor as MlpActor
-        from ..ops import require_extension, functional as Fo
+        from ..ops import require_extension
         if type(actor) is not MlpActor:
             raise TypeError("ActKernel needs the plain MLP Actor")
         if any(l.out_features > 256 for l in actor.layers) or act_dim > 64:

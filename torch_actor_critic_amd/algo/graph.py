@@ -26,7 +26,6 @@ import typing as t
 
 import torch
 
-from ..buffer.replay import Batch
 from ..optim import FlatAdam
 from ..parallel import comm
 from . import sac as sac_mod

@@ -454,3 +454,40 @@ def test_sac_train_bf16_learns_pendulum():
     finally:
         Fo.set_compute_dtype("fp32")
     assert float(np.mean(rets)) > -700.0, rets
+
+
+@pytest.mark.parametrize("hid", [[64], [64, 64, 64]])
+def test_engine_parity_other_depths(hid):
+    """Engine loops are depth-generic: parity at 1 and 3 hidden layers."""
+    global HID
+    old = HID
+    HID = hid
+    try:
+        sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, eng = \
+            _setup()
+        from torch_actor_critic_amd.ops import require_extension
+        ext = require_extension()
+        actor_cpu = copy.deepcopy(actor).cpu()
+        critic_cpu = copy.deepcopy(critic).cpu()
+        target_cpu = copy.deepcopy(target).cpu()
+        torch.manual_seed(7)
+        s = torch.randn(B, O)
+        a = torch.rand(B, A) * 2 - 1
+        r = torch.randn(B)
+        ns = torch.randn(B, O)
+        d = torch.zeros(B)
+        eng.load_batch(s.to(DEV), a.to(DEV), r.to(DEV), ns.to(DEV),
+                       d.to(DEV))
+        eng._run_once()
+        torch.cuda.synchronize()
+        eps = ext.tg_eps(1, 0, 2 * B, A, eng.prob).cpu()
+        ref = _eager_reference(actor_cpu, critic_cpu, target_cpu,
+                               s, a, r, ns, d, eps)
+        cg = q_opt.fp.flat_grad.cpu()
+        ag = pi_opt.fp.flat_grad.cpu()
+        assert torch.allclose(cg, ref["cgrad"], atol=5e-5, rtol=1e-3), \
+            (cg - ref["cgrad"]).abs().max()
+        assert torch.allclose(ag, ref["agrad"], atol=5e-5, rtol=1e-3), \
+            (ag - ref["agrad"]).abs().max()
+    finally:
+        HID = old

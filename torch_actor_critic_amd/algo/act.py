@@ -62,11 +62,7 @@ class ActKernel:
             # host-pinned memory; the host spins on the flag — no H2D/D2H
             # copies, no event sync
             self.flag_pin[0] = 0
-            self.ext.act_step_pinned(
-                self.obs_pin, self.ws, self.bs, self.wmu, self.bmu,
-                self.wls, self.bls, self.act_out, self.act_pin,
-                self.flag_pin, self.ctr, self.seed, self.act_limit,
-                self.lo, self.hi)
+            self.ext.act_fire(self._handle)
             fl = self.flag_np
             for _ in range(2_000_000):
                 if fl[0] != 0:

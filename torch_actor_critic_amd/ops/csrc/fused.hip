@@ -1383,6 +1383,54 @@ void bump3(torch::Tensor a, c10::optional<torch::Tensor> b,
                      c.has_value() ? c->data_ptr<int64_t>() : nullptr);
 }
 
+// Pre-marshalled acting: the ActArgs are built once (act_prepare) and
+// fired with a single-int pybind call per env step (act_fire) — the
+// 13-argument marshalling otherwise costs microseconds per step.
+static std::vector<std::pair<ActArgs, ActPinned>> g_act_handles;
+
+int64_t act_prepare(torch::Tensor x, std::vector<torch::Tensor> ws,
+                    std::vector<torch::Tensor> bs, torch::Tensor wmu,
+                    torch::Tensor bmu, torch::Tensor wls, torch::Tensor bls,
+                    torch::Tensor action, torch::Tensor out_host,
+                    torch::Tensor flag_host, torch::Tensor ctr,
+                    int64_t seed, double act_limit, double lo, double hi) {
+  TORCH_CHECK(out_host.is_pinned() && flag_host.is_pinned());
+  ActArgs a{};
+  a.n_layers = (int)ws.size();
+  TORCH_CHECK(a.n_layers <= ACT_MAXL);
+  a.x = x.data_ptr<float>();
+  a.O = (int)x.numel();
+  a.A = (int)action.numel();
+  int maxw = a.O;
+  for (int L = 0; L < a.n_layers; ++L) {
+    a.w[L] = ws[L].data_ptr<float>();
+    a.b[L] = bs[L].data_ptr<float>();
+    a.width[L] = (int)ws[L].size(0);
+    TORCH_CHECK(a.width[L] <= 256);
+    maxw = std::max(maxw, a.width[L]);
+  }
+  TORCH_CHECK(maxw <= ACT_MAXW && a.A <= 256);
+  a.wmu = wmu.data_ptr<float>();
+  a.bmu = bmu.data_ptr<float>();
+  a.wls = wls.data_ptr<float>();
+  a.bls = bls.data_ptr<float>();
+  a.action = action.data_ptr<float>();
+  a.ctr = ctr.data_ptr<int64_t>();
+  a.seed = (uint64_t)seed;
+  a.act_limit = (float)act_limit;
+  a.lo = (float)lo;
+  a.hi = (float)hi;
+  g_act_handles.push_back({a, ActPinned{out_host.data_ptr<float>(),
+                                        flag_host.data_ptr<int>()}});
+  return (int64_t)g_act_handles.size() - 1;
+}
+
+void act_fire(int64_t handle) {
+  auto& h = g_act_handles.at((size_t)handle);
+  hipLaunchKernelGGL(act_kernel, dim3(1), dim3(256), 0, stream(),
+                     h.first, h.second);
+}
+
 void alpha_update(torch::Tensor log_alpha, torch::Tensor alpha_dev,
                   torch::Tensor m, torch::Tensor v, torch::Tensor step,
                   torch::Tensor mean_logp, double target_entropy, double lr) {
@@ -1411,4 +1459,6 @@ void register_fused(pybind11::module_& m) {
   m.def("act_step", &fused::act_step);
   m.def("bump3", &fused::bump3);
   m.def("act_step_pinned", &fused::act_step_pinned);
+  m.def("act_prepare", &fused::act_prepare);
+  m.def("act_fire", &fused::act_fire);
 }

@@ -51,7 +51,13 @@ class ActKernel:
         self.flag_np = self.flag_pin.numpy()
         self.ev = torch.cuda.Event()
         import os
-        self._pinned_ok = os.environ.get("TAC_AMD_ACT_PINNED", "1") != "0"''
+        self._pinned_ok = os.environ.get("TAC_AMD_ACT_PINNED", "1") != "0"
+        # pre-marshalled launch handle (referenced tensors stay alive via
+        # self.*): one int-arg pybind call per env step
+        self._handle = self.ext.act_prepare(
+            self.obs_pin, self.ws, self.bs, self.wmu, self.bmu, self.wls,
+            self.bls, self.act_out, self.act_pin, self.flag_pin, self.ctr,
+            self.seed, self.act_limit, self.lo, self.hi)
 
     def act(self, state: np.ndarray) -> np.ndarray:
         self.obs_pin.copy_(torch.from_numpy(np.asarray(state,

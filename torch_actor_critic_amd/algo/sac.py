@@ -459,6 +459,11 @@ class SAC:
                 metrics["episode_length"] = float(np.mean(all_lens))
             metrics.update(thr.rates())
             thr.reset()
+            if comm.is_initialized():
+                cs = comm.collective_stats()
+                if cs["allreduce_n"]:
+                    metrics["allreduce_ms_per_call"] = (
+                        1000.0 * cs["allreduce_s"] / cs["allreduce_n"])
 
             if rank0 and logging:
                 if (e + 1) % self.save_every == 0:

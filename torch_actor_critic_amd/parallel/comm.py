@@ -21,10 +21,24 @@ sub-MB payloads):
 
 import datetime
 import os
+import time
 import typing as t
 
 import torch
 import torch.distributed as dist
+
+# cumulative collective timing (per-collective timings are part
+# of the observability surface); read+reset via collective_stats()
+_coll_time = {"allreduce_s": 0.0, "allreduce_n": 0, "broadcast_s": 0.0,
+              "broadcast_n": 0}
+
+
+def collective_stats(reset: bool = True) -> t.Dict[str, float]:
+    out = dict(_coll_time)
+    if reset:
+        for k in _coll_time:
+            _coll_time[k] = 0 if k.endswith("_n") else 0.0
+    return out
 
 
 def is_initialized() -> bool:
@@ -70,7 +84,10 @@ def sync_flat_params(flat: torch.Tensor, src: int = 0):
     """Broadcast a module's whole flat parameter buffer — one collective
     (replaces reference sync_params, sac/mpi.py:93-98)."""
     if is_initialized():
+        t0 = time.perf_counter()
         dist.broadcast(flat, src=src)
+        _coll_time["broadcast_s"] += time.perf_counter() - t0
+        _coll_time["broadcast_n"] += 1
 
 
 def allreduce_grads(flat_grad: torch.Tensor):
@@ -78,8 +95,11 @@ def allreduce_grads(flat_grad: torch.Tensor):
     collective (replaces reference mpi_avg_grads, sac/mpi.py:77-85)."""
     if not is_initialized():
         return
+    t0 = time.perf_counter()
     dist.all_reduce(flat_grad, op=dist.ReduceOp.SUM)
     flat_grad.div_(num_procs())
+    _coll_time["allreduce_s"] += time.perf_counter() - t0
+    _coll_time["allreduce_n"] += 1
 
 
 def allreduce_mean(x: torch.Tensor) -> torch.Tensor:

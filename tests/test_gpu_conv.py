@@ -215,3 +215,38 @@ def test_visual_graphed_update_on_gpu():
     lq, lp = g.read_and_reset_losses(5)
     assert np.isfinite(lq) and np.isfinite(lp)
     assert torch.isfinite(pi_opt.fp.flat).all()
+
+
+def test_paired_double_critic_matches_sequential():
+    """The lockstep (paired-launch) VisualDoubleCritic forward/backward
+    must match running q1/q2 sequentially."""
+    from torch_actor_critic_amd.envs.visual import MultiObservation
+    from torch_actor_critic_amd.models.visual import VisualDoubleCritic
+    torch.manual_seed(9)
+    device = torch.device(DEV)
+    vdc = VisualDoubleCritic(12, 4, (3, 64, 64), [32, 32]).to(device)
+    feats = torch.randn(6, 12, device=device)
+    frames = torch.randn(6, 3, 64, 64, device=device)
+    act = torch.randn(6, 4, device=device)
+    mo = MultiObservation(feats, frames)
+
+    q1p, q2p = vdc(mo, act)                      # paired path
+    q1s = vdc.q1(mo, act)                        # sequential path
+    q2s = vdc.q2(mo, act)
+    assert torch.allclose(q1p, q1s, atol=1e-4), (q1p - q1s).abs().max()
+    assert torch.allclose(q2p, q2s, atol=1e-4)
+
+    # backward parity on a scalar loss
+    loss_p = (q1p.square().mean() + q2p.square().mean())
+    gp = torch.autograd.grad(loss_p, list(vdc.parameters()),
+                             allow_unused=True)
+    q1s2 = vdc.q1(mo, act)
+    q2s2 = vdc.q2(mo, act)
+    loss_s = (q1s2.square().mean() + q2s2.square().mean())
+    gs = torch.autograd.grad(loss_s, list(vdc.parameters()),
+                             allow_unused=True)
+    for a, b in zip(gp, gs):
+        if a is None and b is None:
+            continue
+        assert torch.allclose(a, b, atol=1e-3, rtol=1e-3), \
+            (a - b).abs().max()

@@ -195,4 +195,49 @@ class VisualDoubleCritic(nn.Module):
                                filters, kernel_sizes, strides)
 
     def forward(self, state: MultiObservation, action):
-        return self.q1(state, action), self.q2(state, action)
+        from ..ops import use_native
+        probe = self.q1.layers[0].weight
+        if not use_native(state.features, probe):
+            return self.q1(state, action), self.q2(state, action)
+        return self._forward_paired(state, action)
+
+    def _forward_paired(self, state: MultiObservation, action):
+        """Both critics in lockstep: every identically-shaped layer pair
+        (convs, dense heads, MLP trunk) runs as ONE multi-problem kernel
+        launch (blockIdx.z) — forward and backward."""
+        q1, q2 = self.q1, self.q2
+        image = state.frame
+        if image.ndim == 3:
+            image = image.view((-1, *q1.vis_dim))
+        x = torch.cat([state.features, action], dim=-1)
+        if x.ndim == 1:
+            x = x.view(-1, q1.obs_dim + q1.act_dim)
+
+        # conv trunks (3 paired convs with fused ReLU)
+        v1, v2 = q1.visual_network, q2.visual_network
+        h1 = h2 = image
+        for name in ("conv_0", "conv_1", "conv_2"):
+            c1, c2 = getattr(v1, name), getattr(v2, name)
+            h1, h2 = Fo.conv2d_pair(h1, h2, c1.weight, c1.bias, c2.weight,
+                                    c2.bias, c1.stride[0], relu=True)
+        B = h1.shape[0]
+        h1 = h1.reshape(B, -1)
+        h2 = h2.reshape(B, -1)
+        h1, h2 = Fo.linear_pair(h1, h2, v1.linear.weight, v1.linear.bias,
+                                v2.linear.weight, v2.linear.bias, False)
+        c1out, c2out = Fo.linear_pair(h1, h2, v1.final.weight,
+                                      v1.final.bias, v2.final.weight,
+                                      v2.final.bias, False)
+
+        # MLP trunks — ReLU on ALL layers incl. the final width-1 layer
+        # (reference parity, SURVEY.md Q6)
+        m1 = m2 = x
+        for l1, l2 in zip(q1.layers, q2.layers):
+            m1, m2 = Fo.linear_pair(m1, m2, l1.weight, l1.bias, l2.weight,
+                                    l2.bias, True)
+
+        y1 = torch.cat([m1, c1out], dim=1)
+        y2 = torch.cat([m2, c2out], dim=1)
+        o1, o2 = Fo.linear_pair(y1, y2, q1.final.weight, q1.final.bias,
+                                q2.final.weight, q2.final.bias, False)
+        return torch.squeeze(o1, -1), torch.squeeze(o2, -1)

@@ -125,11 +125,18 @@ DEVINL void kinc(KDec& r, const ConvDims& d) {
 // conv fwd: grid (ceil(M/64), ceil(OC/64)); Y NCHW scatter epilogue
 // ---------------------------------------------------------------------------
 
+struct ConvP {
+  const float* x; const float* w; const float* bias; float* y;
+};
+
 template <bool BF16, bool RELU>
 __global__ __launch_bounds__(256)
-void conv_fwd_kernel(const float* __restrict__ x, const float* __restrict__ w,
-                     const float* __restrict__ bias, float* __restrict__ y,
-                     ConvDims d) {
+void conv_fwd_kernel(ConvP p0, ConvP p1, ConvDims d) {
+  const ConvP& pp = blockIdx.z ? p1 : p0;
+  const float* x = pp.x;
+  const float* w = pp.w;
+  const float* bias = pp.bias;
+  float* y = pp.y;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -222,12 +229,18 @@ void conv_fwd_kernel(const float* __restrict__ x, const float* __restrict__ w,
 // m=(b,iy,ix), k'=(oc,ky,kx).  RELU mask (y>0) applied to dY on gather.
 // ---------------------------------------------------------------------------
 
+struct ConvGP {
+  const float* dy; const float* ymask; const float* wt; float* dx;
+};
+
 template <bool BF16, bool MASK>
 __global__ __launch_bounds__(256)
-void conv_dgrad_kernel(const float* __restrict__ dy,
-                       const float* __restrict__ ymask,
-                       const float* __restrict__ wt,
-                       float* __restrict__ dx, ConvDims d) {
+void conv_dgrad_kernel(ConvGP p0, ConvGP p1, ConvDims d) {
+  const ConvGP& pp = blockIdx.z ? p1 : p0;
+  const float* dy = pp.dy;
+  const float* ymask = pp.ymask;
+  const float* wt = pp.wt;
+  float* dx = pp.dx;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -335,13 +348,22 @@ void conv_dgrad_kernel(const float* __restrict__ dy,
 // Reduction over M; tiles: rows=oc (64), cols=k (64), i-chunks of BK.
 // ---------------------------------------------------------------------------
 
+struct ConvWP {
+  const float* dy; const float* ymask; const float* x; float* part;
+};
+
 template <bool BF16, bool MASK>
 __global__ __launch_bounds__(256)
-void conv_wgrad_kernel(const float* __restrict__ dy,
-                       const float* __restrict__ ymask,
-                       const float* __restrict__ x,
-                       float* __restrict__ part,   // [S][OC*K + OC]
-                       ConvDims d, int m_chunk) {
+void conv_wgrad_kernel(ConvWP p0, ConvWP p1, ConvDims d, int m_chunk,
+                       int split) {
+  // blockIdx.z = slab * nz + z
+  const int nz = (p1.dy != nullptr) ? 2 : 1;
+  const int zz = (int)blockIdx.z % nz;
+  const int slab = (int)blockIdx.z / nz;
+  const ConvWP& pp = zz ? p1 : p0;
+  const float* dy = pp.dy;
+  const float* ymask = pp.ymask;
+  const float* x = pp.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -351,10 +373,11 @@ void conv_wgrad_kernel(const float* __restrict__ dy,
   const int bk0 = blockIdx.y * TB;   // k cols
   const int M = d.B * d.OH * d.OW;
   const int K = d.IC * d.KH * d.KW;
-  const int m_lo = blockIdx.z * m_chunk;
+  const int m_lo = slab * m_chunk;
   const int m_hi = min(M, m_lo + m_chunk);
-  float* dw_p = part + (int64_t)blockIdx.z * ((int64_t)d.OC * K + d.OC);
+  float* dw_p = pp.part + (int64_t)slab * ((int64_t)d.OC * K + d.OC);
   float* db_p = dw_p + (int64_t)d.OC * K;
+  (void)split;
   constexpr int BK = BF16 ? BKB : BKF;
   constexpr int EL = BF16 ? 16 : 4;
   constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
@@ -458,91 +481,143 @@ ConvDims dims_of(const torch::Tensor& x, const torch::Tensor& w, int64_t s) {
   return d;
 }
 
-torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
-                         c10::optional<torch::Tensor> bias, int64_t s,
-                         bool relu) {
-  auto d = dims_of(x, w, s);
-  auto y = torch::empty({d.B, d.OC, d.OH, d.OW}, x.options());
+std::vector<torch::Tensor> conv2d_fwd_multi(
+    std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
+    std::vector<c10::optional<torch::Tensor>> biases, int64_t s,
+    bool relu) {
+  const int nz = (int)xs.size();
+  TORCH_CHECK(nz >= 1 && nz <= 2);
+  auto d = dims_of(xs[0], ws[0], s);
+  std::vector<torch::Tensor> ys;
+  ConvP p[2] = {};
+  for (int z = 0; z < nz; ++z) {
+    ys.push_back(torch::empty({d.B, d.OC, d.OH, d.OW}, xs[z].options()));
+    p[z] = ConvP{xs[z].data_ptr<float>(), ws[z].data_ptr<float>(),
+                 biases[z].has_value() ? biases[z]->data_ptr<float>()
+                                       : nullptr,
+                 ys[z].data_ptr<float>()};
+  }
   const int M = d.B * d.OH * d.OW;
-  dim3 grid((M + TB - 1) / TB, (d.OC + TB - 1) / TB);
-  const float* bp = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  dim3 grid((M + TB - 1) / TB, (d.OC + TB - 1) / TB, nz);
   const bool bf16 = *g_bf16_flag2;
   auto L = [&](auto b16, auto rl) {
     hipLaunchKernelGGL((conv_fwd_kernel<decltype(b16)::value,
                                         decltype(rl)::value>),
-                       grid, dim3(256), 0, stream(), x.data_ptr<float>(),
-                       w.data_ptr<float>(), bp, y.data_ptr<float>(), d);
+                       grid, dim3(256), 0, stream(), p[0], p[1], d);
   };
   if (bf16) { if (relu) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }
   else      { if (relu) L(std::false_type{}, std::true_type{});
               else L(std::false_type{}, std::false_type{}); }
-  return y;
+  return ys;
+}
+
+torch::Tensor conv2d_fwd(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, int64_t s,
+                         bool relu) {
+  return conv2d_fwd_multi({x}, {w}, {bias}, s, relu)[0];
+}
+
+std::vector<torch::Tensor> conv2d_dgrad_multi(
+    std::vector<torch::Tensor> dys,
+    std::vector<c10::optional<torch::Tensor>> ymasks,
+    std::vector<torch::Tensor> wts, torch::Tensor x_like,
+    torch::Tensor w, int64_t s) {
+  const int nz = (int)dys.size();
+  TORCH_CHECK(nz >= 1 && nz <= 2);
+  auto d = dims_of(x_like, w, s);
+  std::vector<torch::Tensor> dxs;
+  ConvGP p[2] = {};
+  const bool mask = ymasks[0].has_value();
+  for (int z = 0; z < nz; ++z) {
+    dxs.push_back(torch::empty_like(x_like));
+    p[z] = ConvGP{dys[z].data_ptr<float>(),
+                  ymasks[z].has_value() ? ymasks[z]->data_ptr<float>()
+                                        : nullptr,
+                  wts[z].data_ptr<float>(), dxs[z].data_ptr<float>()};
+  }
+  const int M = d.B * d.IH * d.IW;
+  dim3 grid((M + TB - 1) / TB, (d.IC + TB - 1) / TB, nz);
+  const bool bf16 = *g_bf16_flag2;
+  auto L = [&](auto b16, auto mk) {
+    hipLaunchKernelGGL((conv_dgrad_kernel<decltype(b16)::value,
+                                          decltype(mk)::value>),
+                       grid, dim3(256), 0, stream(), p[0], p[1], d);
+  };
+  if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
+              else L(std::true_type{}, std::false_type{}); }
+  else      { if (mask) L(std::false_type{}, std::true_type{});
+              else L(std::false_type{}, std::false_type{}); }
+  return dxs;
 }
 
 torch::Tensor conv2d_dgrad(torch::Tensor dy,
                            c10::optional<torch::Tensor> ymask,
                            torch::Tensor wt, torch::Tensor x_like,
                            torch::Tensor w, int64_t s) {
-  auto d = dims_of(x_like, w, s);
-  auto dx = torch::empty_like(x_like);
-  const int M = d.B * d.IH * d.IW;
-  dim3 grid((M + TB - 1) / TB, (d.IC + TB - 1) / TB);
-  const bool mask = ymask.has_value();
-  const float* mp = mask ? ymask->data_ptr<float>() : nullptr;
-  const bool bf16 = *g_bf16_flag2;
-  auto L = [&](auto b16, auto mk) {
-    hipLaunchKernelGGL((conv_dgrad_kernel<decltype(b16)::value,
-                                          decltype(mk)::value>),
-                       grid, dim3(256), 0, stream(), dy.data_ptr<float>(),
-                       mp, wt.data_ptr<float>(), dx.data_ptr<float>(), d);
-  };
-  if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
-              else L(std::true_type{}, std::false_type{}); }
-  else      { if (mask) L(std::false_type{}, std::true_type{});
-              else L(std::false_type{}, std::false_type{}); }
-  return dx;
+  return conv2d_dgrad_multi({dy}, {ymask}, {wt}, x_like, w, s)[0];
 }
 
-std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
-                                        c10::optional<torch::Tensor> ymask,
-                                        torch::Tensor x, torch::Tensor w,
-                                        int64_t s) {
-  auto d = dims_of(x, w, s);
-  auto dw = torch::empty_like(w);
-  auto db = torch::empty({d.OC}, w.options());
+std::vector<torch::Tensor> conv2d_wgrad_multi(
+    std::vector<torch::Tensor> dys,
+    std::vector<c10::optional<torch::Tensor>> ymasks,
+    std::vector<torch::Tensor> xs, torch::Tensor w, int64_t s) {
+  const int nz = (int)dys.size();
+  TORCH_CHECK(nz >= 1 && nz <= 2);
+  auto d = dims_of(xs[0], w, s);
   const int K = d.IC * d.KH * d.KW;
   const int M = d.B * d.OH * d.OW;
-  const int tiles = ((d.OC + TB - 1) / TB) * ((K + TB - 1) / TB);
-  // split the batch reduction so ~256+ blocks are in flight
+  const int tiles = ((d.OC + TB - 1) / TB) * ((K + TB - 1) / TB) * nz;
   const int BKc = 64;
   int max_split = (M + BKc - 1) / BKc;
   int split = std::max(1, std::min({max_split,
                                     (256 + tiles - 1) / tiles, 64}));
   int m_chunk = ((M + split - 1) / split + BKc - 1) / BKc * BKc;
   split = (M + m_chunk - 1) / m_chunk;
-  auto part = torch::empty({split, (int64_t)d.OC * K + d.OC}, w.options());
-  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB, split);
-  const bool mask = ymask.has_value();
-  const float* mp = mask ? ymask->data_ptr<float>() : nullptr;
+  const int64_t per = (int64_t)d.OC * K + d.OC;
+  const bool mask = ymasks[0].has_value();
+  std::vector<torch::Tensor> outs;   // dw0, db0[, dw1, db1]
+  std::vector<torch::Tensor> parts;
+  ConvWP p[2] = {};
+  for (int z = 0; z < nz; ++z) {
+    outs.push_back(torch::empty_like(w));
+    outs.push_back(torch::empty({d.OC}, w.options()));
+    parts.push_back(torch::empty({split, per}, w.options()));
+    p[z] = ConvWP{dys[z].data_ptr<float>(),
+                  ymasks[z].has_value() ? ymasks[z]->data_ptr<float>()
+                                        : nullptr,
+                  xs[z].data_ptr<float>(), parts[z].data_ptr<float>()};
+  }
+  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB, nz * split);
   const bool bf16 = *g_bf16_flag2;
   auto L = [&](auto b16, auto mk) {
     hipLaunchKernelGGL((conv_wgrad_kernel<decltype(b16)::value,
                                           decltype(mk)::value>),
-                       grid, dim3(256), 0, stream(), dy.data_ptr<float>(),
-                       mp, x.data_ptr<float>(), part.data_ptr<float>(), d,
-                       m_chunk);
+                       grid, dim3(256), 0, stream(), p[0], p[1], d,
+                       m_chunk, split);
   };
   if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }
   else      { if (mask) L(std::false_type{}, std::true_type{});
               else L(std::false_type{}, std::false_type{}); }
   int64_t dw_n = (int64_t)d.OC * K;
-  int blocks = (int)std::min<int64_t>((dw_n + d.OC + 255) / 256, 512);
-  hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
-                     stream(), part.data_ptr<float>(), dw.data_ptr<float>(),
-                     db.data_ptr<float>(), dw_n, (int64_t)d.OC, split);
-  return {dw, db};
+  int blocks = (int)std::min<int64_t>((per + 255) / 256, 512);
+  for (int z = 0; z < nz; ++z) {
+    hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
+                       stream(), parts[z].data_ptr<float>(),
+                       outs[2 * z].data_ptr<float>(),
+                       outs[2 * z + 1].data_ptr<float>(), dw_n,
+                       (int64_t)d.OC, split);
+  }
+  return outs;
+}
+
+std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
+                                        c10::optional<torch::Tensor> ymask,
+                                        torch::Tensor x, torch::Tensor w,
+                                        int64_t s) {
+  auto o = conv2d_wgrad_multi({dy}, {ymask}, {x}, w, s);
+  return {o[0], o[1]};
 }
 
 }  // namespace convk
@@ -555,4 +630,7 @@ void register_conv(pybind11::module_& m) {
   m.def("conv2d_fwd", &convk::conv2d_fwd);
   m.def("conv2d_dgrad", &convk::conv2d_dgrad);
   m.def("conv2d_wgrad", &convk::conv2d_wgrad);
+  m.def("conv2d_fwd_multi", &convk::conv2d_fwd_multi);
+  m.def("conv2d_dgrad_multi", &convk::conv2d_dgrad_multi);
+  m.def("conv2d_wgrad_multi", &convk::conv2d_wgrad_multi);
 }

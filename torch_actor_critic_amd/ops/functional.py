@@ -178,6 +178,124 @@ def conv2d(x, w, b, stride: int, relu: bool = False):
 
 
 # ---------------------------------------------------------------------------
+# Paired twin-critic ops: both critics' identically-shaped layers run in
+# ONE launch (blockIdx.z), forward and backward (the visual DoubleCritic
+#runs lockstep through these)
+# ---------------------------------------------------------------------------
+
+class _PairedLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x1, x2, w1, b1, w2, b2, relu):
+        ext = require_extension()
+        x1 = x1.contiguous()
+        x2 = x2.contiguous()
+        M, K = x1.shape
+        N = w1.shape[0]
+        y1 = torch.empty(M, N, device=x1.device, dtype=x1.dtype)
+        y2 = torch.empty_like(y1)
+        ext.mgemm([x1, x2], [w1, w2], [b1, b2], [y1, y2], [None, None],
+                  M, N, K, K, N, relu, [], [], [], 0, 0, 0, [])
+        ctx.save_for_backward(x1, x2, w1, w2, y1, y2)
+        ctx.relu = relu
+        return y1, y2
+
+    @staticmethod
+    def backward(ctx, dy1, dy2):
+        x1, x2, w1, w2, y1, y2 = ctx.saved_tensors
+        ext = require_extension()
+        dy1 = dy1.contiguous()
+        dy2 = dy2.contiguous()
+        M, K = x1.shape
+        N = w1.shape[0]
+        m1 = y1 if ctx.relu else None
+        m2 = y2 if ctx.relu else None
+        dw1 = torch.empty_like(w1)
+        dw2 = torch.empty_like(w2)
+        db1 = torch.empty(N, device=w1.device, dtype=w1.dtype)
+        db2 = torch.empty_like(db1)
+        ext.mwgrad([dy1, dy2], [m1, m2], [x1, x2], [dw1, dw2],
+                   [db1, db2], M, N, K, N, K, 0)
+        dx1 = dx2 = None
+        if ctx.needs_input_grad[0]:
+            wt1 = torch.empty(K, N, device=w1.device, dtype=w1.dtype)
+            wt2 = torch.empty_like(wt1)
+            ext.transpose_multi([w1, w2], [wt1, wt2])
+            dx1 = torch.empty_like(x1)
+            dx2 = torch.empty_like(x2)
+            ext.mgemm([dy1, dy2], [wt1, wt2], [None, None], [dx1, dx2],
+                      [m1, m2], M, K, N, N, K, False, [], [], [], 0, 0, 0,
+                      [])
+        return dx1, dx2, dw1, db1, dw2, db2, None
+
+
+def linear_pair(x1, x2, w1, b1, w2, b2, relu: bool):
+    if use_native(x1, w1):
+        return _PairedLinear.apply(x1, x2, w1, b1, w2, b2, relu)
+    y1 = F.linear(x1, w1, b1)
+    y2 = F.linear(x2, w2, b2)
+    if relu:
+        y1, y2 = F.relu(y1), F.relu(y2)
+    return y1, y2
+
+
+class _PairedConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x1, x2, w1, b1, w2, b2, stride, relu):
+        ext = require_extension()
+        y1, y2 = ext.conv2d_fwd_multi(
+            [x1.contiguous(), x2.contiguous()],
+            [w1.contiguous(), w2.contiguous()],
+            [b1.contiguous() if b1 is not None else None,
+             b2.contiguous() if b2 is not None else None], stride, relu)
+        if relu:
+            ctx.save_for_backward(x1, x2, w1, w2, y1, y2)
+        else:
+            ctx.save_for_backward(x1, x2, w1, w2)
+        ctx.stride = stride
+        ctx.relu = relu
+        ctx.has_bias = b1 is not None
+        return y1, y2
+
+    @staticmethod
+    def backward(ctx, dy1, dy2):
+        ext = require_extension()
+        dy1 = dy1.contiguous()
+        dy2 = dy2.contiguous()
+        if ctx.relu:
+            x1, x2, w1, w2, y1, y2 = ctx.saved_tensors
+            dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
+            dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
+        else:
+            x1, x2, w1, w2 = ctx.saved_tensors
+        dx1 = dx2 = None
+        if ctx.needs_input_grad[0]:
+            oc, ic, kh, kw = w1.shape
+            wt1 = w1.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw
+                                                 ).contiguous()
+            wt2 = w2.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw
+                                                 ).contiguous()
+            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [None, None],
+                                              [wt1, wt2], x1, w1,
+                                              ctx.stride)
+        dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
+            [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
+        if not ctx.has_bias:
+            db1 = db2 = None
+        return dx1, dx2, dw1, db1, dw2, db2, None, None
+
+
+def conv2d_pair(x1, x2, w1, b1, w2, b2, stride: int, relu: bool):
+    if use_native(x1, w1):
+        return _PairedConv2d.apply(x1, x2, w1, b1, w2, b2, int(stride),
+                                   relu)
+    y1 = F.conv2d(x1, w1, b1, stride=stride)
+    y2 = F.conv2d(x2, w2, b2, stride=stride)
+    if relu:
+        y1, y2 = F.relu(y1), F.relu(y2)
+    return y1, y2
+
+
+# ---------------------------------------------------------------------------
 # Fused tanh-Gaussian head (sample + squash + log-prob)
 # ---------------------------------------------------------------------------
 

@@ -27,6 +27,7 @@ update order, losses and polyak follow sac/algorithm.py:115-162,77-81.
 """
 
 import math
+import os
 import typing as t
 
 import torch
@@ -56,6 +57,10 @@ class FusedSACEngine:
         self.device = device
         self.sample = sample
         self.world = comm.num_procs()
+        # TAC_AMD_GRAPH_COLL=force exercises the in-graph-collective
+        # capture on a 1-rank RCCL communicator (single-GPU testable)
+        self._force_coll = (os.environ.get("TAC_AMD_GRAPH_COLL") == "force"
+                            and comm.is_initialized())
         self.seed = philox_seed
 
         B = self.B = batch_size
@@ -483,9 +488,12 @@ class FusedSACEngine:
 
     # ------------------------------------------------------------------
 
-    def _reduce(self, opt):
-        if self.world > 1:
-            comm.allreduce_grads(opt.fp.flat_grad)
+    def _reduce(self, opt, in_graph: bool = False):
+        if self.world > 1 or self._force_coll:
+            if in_graph:
+                comm.allreduce_grads_capturable(opt.fp.flat_grad)
+            else:
+                comm.allreduce_grads(opt.fp.flat_grad)
 
     def _run_once(self):
         self._phase_critic()
@@ -495,7 +503,6 @@ class FusedSACEngine:
         self._phase_finish()
 
     def _capture(self):
-        import os
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -506,8 +513,34 @@ class FusedSACEngine:
         # TAC_AMD_SPLIT_GRAPHS=1 forces the data-parallel 3-graph
         # structure at world=1 so the multi-GPU capture path is testable
         # on a single GPU
-        split = (self.world > 1
+        split = (self.world > 1 or self._force_coll
                  or os.environ.get("TAC_AMD_SPLIT_GRAPHS") == "1")
+        # Default data-parallel fast path: record BOTH flat-bucket RCCL
+        # all-reduces inside ONE hipGraph — one replay per update, no
+        # host round-trips between the three phases (xGMI collectives
+        # at these sub-MB payloads are latency-bound; so is the host).
+        # Falls back to 3 graphs + host-issued collectives if capture
+        # of the communicator is refused.
+        coll_in_graph = ((self.world > 1 or self._force_coll)
+                         and comm.backend_name() == "nccl"
+                         and os.environ.get("TAC_AMD_GRAPH_COLL", "1")
+                         != "0")
+        if coll_in_graph:
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._phase_critic()
+                    self._reduce(self.q_opt, in_graph=True)
+                    self._phase_policy()
+                    self._reduce(self.pi_opt, in_graph=True)
+                    self._phase_finish()
+                self.graph = g
+                return
+            except Exception as e:  # pragma: no cover - fallback path
+                import logging
+                logging.getLogger(__name__).warning(
+                    "in-graph collective capture failed (%s); "
+                    "falling back to split graphs", e)
         if not split:
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):

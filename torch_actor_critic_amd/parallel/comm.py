@@ -60,6 +60,13 @@ def init_distributed(backend: t.Optional[str] = None) -> t.Tuple[int, int]:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
+        if backend == "nccl" and \
+                os.environ.get("TAC_AMD_GRAPH_COLL", "1") != "0":
+            # in-graph RCCL collectives (the default data-parallel fast
+            # path) require the watchdog's async error handling off so
+            # captured works are not event-queried by the watchdog
+            os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
+            os.environ.setdefault("NCCL_ASYNC_ERROR_HANDLING", "0")
         dist.init_process_group(backend=backend,
                                 timeout=datetime.timedelta(seconds=300))
         if backend == "nccl":
@@ -100,6 +107,20 @@ def allreduce_grads(flat_grad: torch.Tensor):
     flat_grad.div_(num_procs())
     _coll_time["allreduce_s"] += time.perf_counter() - t0
     _coll_time["allreduce_n"] += 1
+
+
+def backend_name() -> str:
+    return str(dist.get_backend()) if is_initialized() else ""
+
+
+def allreduce_grads_capturable(flat_grad: torch.Tensor):
+    """allreduce_grads without host-side timing/bookkeeping — safe to
+    record inside a hipGraph capture (RCCL supports captured
+    collectives; the div-by-world kernel is captured too).  Host timers
+    are meaningless for replayed collectives, so the per-collective
+    counters deliberately see only host-issued calls."""
+    dist.all_reduce(flat_grad, op=dist.ReduceOp.SUM)
+    flat_grad.div_(num_procs())
 
 
 def allreduce_mean(x: torch.Tensor) -> torch.Tensor:

@@ -143,8 +143,7 @@ class FusedSACEngine:
         # SLOWER (A/B 4344 vs 4613 upd/s at batch 64) — the single-block
         # outer-product write loses to the 8-block GEMM it replaces.
         # Kept for study via TAC_AMD_LOSS_FUSE=1; default off.
-        import os as _os
-        self._loss_fuse = (_os.environ.get("TAC_AMD_LOSS_FUSE") == "1"
+        self._loss_fuse = (os.environ.get("TAC_AMD_LOSS_FUSE") == "1"
                            and B <= 1024 and len(self.c_w) >= 2)
 
         # whole-MLP fused forward feasibility (LDS budget)
@@ -156,7 +155,6 @@ class FusedSACEngine:
         # (683 vs 777): the layered path's N/M-parallel grid beats the
         # fused kernel's per-block layer serialization.  Kept available
         # for study via TAC_AMD_MLPF=1; default off.
-        import os
         self.use_mlpf = (
             os.environ.get("TAC_AMD_MLPF") == "1"
             and self.ext.mlp_fwd_fits(OC, list(self.c_w), bf16)

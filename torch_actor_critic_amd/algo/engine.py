@@ -186,7 +186,7 @@ class FusedSACEngine:
         # 5464 updates/s — the cross-queue dependency overhead dwarfs the
         # ~5 us kernels at this scale.  Kept for study via
         # TAC_AMD_WGRAD_STREAM=1; default off.
-        self._use_side = _os.environ.get("TAC_AMD_WGRAD_STREAM") == "1"
+        self._use_side = os.environ.get("TAC_AMD_WGRAD_STREAM") == "1"
         self._s2 = torch.cuda.Stream()
         self._fork_evs = [torch.cuda.Event() for _ in range(12)]
         self._join_evs = [torch.cuda.Event() for _ in range(4)]

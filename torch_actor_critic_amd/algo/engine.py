@@ -546,16 +546,23 @@ class FusedSACEngine:
                 self._phase_policy()
                 self._phase_finish()
         else:
-            g1 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g1):
-                self._phase_critic()
-            g2 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g2, pool=g1.pool()):
-                self._phase_policy()
-            g3 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g3, pool=g1.pool()):
-                self._phase_finish()
-            self._graphs = (g1, g2, g3)
+            try:
+                g1 = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g1):
+                    self._phase_critic()
+                g2 = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g2, pool=g1.pool()):
+                    self._phase_policy()
+                g3 = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g3, pool=g1.pool()):
+                    self._phase_finish()
+                self._graphs = (g1, g2, g3)
+            except Exception as e:  # pragma: no cover - last-resort path
+                import logging
+                logging.getLogger(__name__).warning(
+                    "split-graph capture failed (%s); running the fused "
+                    "schedule uncaptured", e)
+                self._graphs = None
 
     def step(self):
         if self.graph is not None:

@@ -136,6 +136,8 @@ def gather_stats(values: t.Sequence[float], device=None) -> t.List[float]:
     if not is_initialized():
         return list(values)
     world = num_procs()
+    if device is None and backend_name() == "nccl":
+        device = torch.device("cuda", torch.cuda.current_device())
     n = torch.tensor([len(values)], dtype=torch.int64)
     if device is not None:
         n = n.to(device)

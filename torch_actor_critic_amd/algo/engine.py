@@ -187,6 +187,13 @@ class FusedSACEngine:
         # ~5 us kernels at this scale.  Kept for study via
         # TAC_AMD_WGRAD_STREAM=1; default off.
         self._use_side = os.environ.get("TAC_AMD_WGRAD_STREAM") == "1"
+        # Batch every wgrad of a backward phase into ONE heterogeneous
+        # multi-problem launch (launch-latency win at small batch; the
+        # split-M path is better at large batch, so gate on B).
+        # TAC_AMD_WGRAD_BATCH=0 disables for A/B.
+        self._wgrad_batch = (os.environ.get("TAC_AMD_WGRAD_BATCH", "1")
+                             == "1" and B <= 256 and not self._use_side)
+        self._wjobs: t.List[tuple] = []
         self._s2 = torch.cuda.Stream()
         self._fork_evs = [torch.cuda.Event() for _ in range(12)]
         self._join_evs = [torch.cuda.Event() for _ in range(4)]
@@ -227,6 +234,26 @@ class FusedSACEngine:
         with torch.cuda.stream(self._s2):
             ev.record()
         torch.cuda.current_stream().wait_event(ev)
+
+    def _wgrad(self, dys, masks, xs, dws, dbs, M, N, K, lddy, ldx, xoff):
+        """Route one wgrad problem group: queue for the phase-wide
+        heterogeneous launch, or issue the per-layer mwgrad directly."""
+        if self._wgrad_batch:
+            for i in range(len(dys)):
+                self._wjobs.append((dys[i], masks[i], xs[i], dws[i],
+                                    dbs[i], M, N, K, lddy, ldx, xoff))
+        else:
+            self._side(lambda: self.ext.mwgrad(
+                list(dys), list(masks), list(xs), list(dws), list(dbs),
+                M, N, K, lddy, ldx, xoff))
+
+    def _wflush(self):
+        """Issue all queued wgrad problems of this phase in ONE launch."""
+        if not self._wjobs:
+            return
+        cols = list(zip(*self._wjobs))
+        self.ext.mwgrad_het(*[list(c) for c in cols])
+        self._wjobs = []
 
     def _mg(self, xs, ws, bs, ys, masks, M, N, K, lda, ldy, relu,
             xs2=None, ws2=None, masks2=None, K2=0, x_off=0, x2_off=0,
@@ -370,15 +397,12 @@ class FusedSACEngine:
             else:
                 x_in = [self.XC, self.XC]
                 ldx, xoff = OC, 0
-            dd, mm, xx = list(d), list(masks), list(x_in)
-            ii = i
-            self._side(lambda: ext.mwgrad(
-                dd, mm, xx,
-                [self.cw[z][ii][0].grad for z in range(2)],
-                [self.cw[z][ii][1].grad for z in range(2)],
-                B, self.c_w[ii],
-                (self.c_w[ii - 1] if ii > 0 else OC),
-                self.c_w[ii], ldx, xoff))
+            self._wgrad(list(d), list(masks), list(x_in),
+                        [self.cw[z][i][0].grad for z in range(2)],
+                        [self.cw[z][i][1].grad for z in range(2)],
+                        B, self.c_w[i],
+                        (self.c_w[i - 1] if i > 0 else OC),
+                        self.c_w[i], ldx, xoff)
             if i > 0:
                 if i == nL - 1 and fuse:
                     pass  # dy2 already produced by the fused loss kernel
@@ -390,6 +414,7 @@ class FusedSACEngine:
                              self.c_w[i], self.c_w[i - 1], False)
                 d = [self.dc[z][i - 1] for z in range(2)]
         # critic grads must be complete before all-reduce / Adam
+        self._wflush()
         self._join_side()
 
     def _phase_policy(self):
@@ -399,7 +424,8 @@ class FusedSACEngine:
         qo = self.q_opt
         ext.adam_t(qo.fp.flat, qo.fp.flat_grad, qo.m, qo.v, qo.step_t,
                    qo.lr, qo.betas[0], qo.betas[1], qo.eps,
-                   qo.weight_decay, self._c_offs, self._c_tr_dst)
+                   qo.weight_decay, self._c_offs, self._c_tr_dst,
+                   self.target_flat, self.sac.polyak)
 
         # critic on (s, pi) = XC2 with the UPDATED critic
         self._critic_fwd(self.XC2, 0, self.cw, self.p_act, OC)
@@ -442,10 +468,9 @@ class FusedSACEngine:
         h_last = self.a_hidden[-1]
         a_last = self.a_act[-1]
         (wm, bm), (wl, bl) = self.head_w
-        self._side(lambda: ext.mwgrad(
-            [self.dmu, self.dls], [None, None], [a_last, a_last],
-            [wm.grad, wl.grad], [bm.grad, bl.grad],
-            B, A, h_last, A, h_last, 0))
+        self._wgrad([self.dmu, self.dls], [None, None], [a_last, a_last],
+                    [wm.grad, wl.grad], [bm.grad, bl.grad],
+                    B, A, h_last, A, h_last, 0)
         self._mg([self.dmu], [self.hwt[0]], [None], [self.da[-1]], [None],
                  B, h_last, A, A, h_last, False,
                  xs2=[self.dls], ws2=[self.hwt[1]], masks2=[None], K2=A)
@@ -460,16 +485,16 @@ class FusedSACEngine:
                 x_in, ldx, xoff = self.XC, OC, 0
                 kin = O
             (w, b) = self.aw[i]
-            dd, mm2, xx2, ii = d, mask, x_in, i
-            self._side(lambda: ext.mwgrad(
-                [dd], [mm2], [xx2], [w.grad], [b.grad],
-                B, self.a_hidden[ii], kin, self.a_hidden[ii], ldx, xoff))
+            self._wgrad([d], [mask], [x_in], [w.grad], [b.grad],
+                        B, self.a_hidden[i], kin, self.a_hidden[i],
+                        ldx, xoff)
             if i > 0:
                 self._mg([d], [self.awt[i]], [None], [self.da[i - 1]],
                          [mask], B, self.a_hidden[i - 1], self.a_hidden[i],
                          self.a_hidden[i], self.a_hidden[i - 1], False)
                 d = self.da[i - 1]
         # actor grads must be complete before all-reduce / Adam
+        self._wflush()
         self._join_side()
 
     def _phase_finish(self):
@@ -477,8 +502,9 @@ class FusedSACEngine:
         po = self.pi_opt
         ext.adam_t(po.fp.flat, po.fp.flat_grad, po.m, po.v, po.step_t,
                    po.lr, po.betas[0], po.betas[1], po.eps,
-                   po.weight_decay, self._a_offs, self._a_tr_dst)
-        ext.polyak_(self.target_flat, self.q_opt.fp.flat, self.sac.polyak)
+                   po.weight_decay, self._a_offs, self._a_tr_dst,
+                   None, 0.0)
+        # polyak is fused into the critic adam_t (see _phase_policy)
         if self.learn_alpha:
             ext.alpha_update(self.log_alpha, self.alpha_dev, self.alpha_m,
                              self.alpha_v, self.alpha_step, self.mean_logp,

@@ -402,32 +402,21 @@ struct WGemm {
 };
 
 template <bool BF16, bool MASK>
-__global__ __launch_bounds__(256)
-void mwgrad_kernel(WGemm g) {
-  const int zz = (int)blockIdx.z % g.nz;
-  const int slab = (int)blockIdx.z / g.nz;
-  const WProb& p = g.p[zz];
-  const int m_lo = slab * g.m_chunk;
-  const int m_hi = min(g.M, m_lo + g.m_chunk);
-  float* dw_out = p.dw;
-  float* db_out = p.db;
-  if (g.part) {
-    float* base = g.part
-        + ((int64_t)slab * g.nz + zz) * ((int64_t)g.N * g.K + g.N);
-    dw_out = base;
-    db_out = p.db ? base + (int64_t)g.N * g.K : nullptr;
-  }
+DEVINL void wgrad_tile_body(const float* __restrict__ dy,
+                            const float* __restrict__ ymask,
+                            const float* __restrict__ x,
+                            float* __restrict__ dw_out,
+                            float* __restrict__ db_out, bool has_db,
+                            int M, int N, int K, int lddy, int ldx,
+                            int m_lo, int m_hi, int bn0, int bk0,
+                            char* smem, float* dbs) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int wrow = (wid >> 1) * 32;
   const int wcol = (wid & 1) * 32;
-  const int bn0 = blockIdx.x * TB;   // output rows (N of dW)
-  const int bk0 = blockIdx.y * TB;   // output cols (K of dW)
   constexpr int BK = BF16 ? BKB2 : BKF2;
   constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
-  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
-  __shared__ float dbs[64];
 
   if (tid < 64) dbs[tid] = 0.f;
   f32x4 acc[2][2] = {};
@@ -441,18 +430,18 @@ void mwgrad_kernel(WGemm g) {
         const int ic = tid & 63;
         const int nc0 = (tid >> 6) * 16;
         const int gi = i0 + ic;
-        bool interior = (i0 + 64 <= m_hi) && (bn0 + 64 <= g.N)
-                        && ((g.lddy & 3) == 0) && ((bn0 & 3) == 0);
+        bool interior = (i0 + 64 <= m_hi) && (bn0 + 64 <= N)
+                        && ((lddy & 3) == 0) && ((bn0 & 3) == 0);
         float v[16];
         if (interior) {
-          const float* q = p.dy + (int64_t)gi * g.lddy + bn0 + nc0;
+          const float* q = dy + (int64_t)gi * lddy + bn0 + nc0;
 #pragma unroll
           for (int t = 0; t < 4; ++t) {
             float4 f = *(const float4*)(q + t * 4);
             v[t*4+0]=f.x; v[t*4+1]=f.y; v[t*4+2]=f.z; v[t*4+3]=f.w;
           }
           if constexpr (MASK) {
-            const float* mq = p.ymask + (int64_t)gi * g.lddy + bn0 + nc0;
+            const float* mq = ymask + (int64_t)gi * lddy + bn0 + nc0;
 #pragma unroll
             for (int t = 0; t < 4; ++t) {
               float4 f = *(const float4*)(mq + t * 4);
@@ -467,10 +456,10 @@ void mwgrad_kernel(WGemm g) {
           for (int e = 0; e < 16; ++e) {
             int n = bn0 + nc0 + e;
             float val = 0.f;
-            if (gi < m_hi && n < g.N) {
-              val = p.dy[(int64_t)gi * g.lddy + n];
+            if (gi < m_hi && n < N) {
+              val = dy[(int64_t)gi * lddy + n];
               if constexpr (MASK) {
-                val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
+                val = ymask[(int64_t)gi * lddy + n] > 0.f ? val : 0.f;
               }
             }
             v[e] = val;
@@ -493,10 +482,10 @@ void mwgrad_kernel(WGemm g) {
           int n = bn0 + nc0 + e;
           int gi = i0 + ic;
           float val = 0.f;
-          if (gi < m_hi && n < g.N) {
-            val = p.dy[(int64_t)gi * g.lddy + n];
+          if (gi < m_hi && n < N) {
+            val = dy[(int64_t)gi * lddy + n];
             if constexpr (MASK) {
-              val = p.ymask[(int64_t)gi * g.lddy + n] > 0.f ? val : 0.f;
+              val = ymask[(int64_t)gi * lddy + n] > 0.f ? val : 0.f;
             }
           }
           as[(nc0 + e) * LDSF2 + ic] = val;
@@ -510,11 +499,11 @@ void mwgrad_kernel(WGemm g) {
         const int ic = tid & 63;
         const int kc0 = (tid >> 6) * 16;
         const int gi = i0 + ic;
-        bool interior = (i0 + 64 <= m_hi) && (bk0 + 64 <= g.K)
-                        && ((g.ldx & 3) == 0) && ((bk0 & 3) == 0);
+        bool interior = (i0 + 64 <= m_hi) && (bk0 + 64 <= K)
+                        && ((ldx & 3) == 0) && ((bk0 & 3) == 0);
         float v[16];
         if (interior) {
-          const float* q = p.x + (int64_t)gi * g.ldx + bk0 + kc0;
+          const float* q = x + (int64_t)gi * ldx + bk0 + kc0;
 #pragma unroll
           for (int t = 0; t < 4; ++t) {
             float4 f = *(const float4*)(q + t * 4);
@@ -525,8 +514,8 @@ void mwgrad_kernel(WGemm g) {
           for (int e = 0; e < 16; ++e) {
             int k = bk0 + kc0 + e;
             int gi2 = i0 + ic;
-            v[e] = (gi2 < m_hi && k < g.K)
-                       ? p.x[(int64_t)gi2 * g.ldx + k] : 0.f;
+            v[e] = (gi2 < m_hi && k < K)
+                       ? x[(int64_t)gi2 * ldx + k] : 0.f;
           }
         }
 #pragma unroll
@@ -541,14 +530,14 @@ void mwgrad_kernel(WGemm g) {
           int k = bk0 + kc0 + e;
           int gi = i0 + ic;
           bs[(kc0 + e) * LDSF2 + ic] =
-              (gi < m_hi && k < g.K) ? p.x[(int64_t)gi * g.ldx + k] : 0.f;
+              (gi < m_hi && k < K) ? x[(int64_t)gi * ldx + k] : 0.f;
         }
       }
     }
     __syncthreads();
     mma_tiles<BF16>(smem, smem + LBYTES, acc, lane, wrow, wcol);
     // db: from the staged (already masked) A tile, k-tile-0 blocks only
-    if (p.db && blockIdx.y == 0 && tid < 64) {
+    if (has_db && bk0 == 0 && tid < 64) {
       float s = 0.f;
       if constexpr (BF16) {
         const __bf16* as = (const __bf16*)smem;
@@ -572,11 +561,79 @@ void mwgrad_kernel(WGemm g) {
       for (int r = 0; r < 4; ++r) {
         int gn = bn0 + wrow + mi * 16 + crow + r;
         int gk = bk0 + wcol + ni * 16 + ccol;
-        if (gn < g.N && gk < g.K)
-          dw_out[(int64_t)gn * g.K + gk] = acc[mi][ni][r];
+        if (gn < N && gk < K)
+          dw_out[(int64_t)gn * K + gk] = acc[mi][ni][r];
       }
-  if (db_out && blockIdx.y == 0 && tid < 64 && bn0 + tid < g.N)
+  if (db_out && bk0 == 0 && tid < 64 && bn0 + tid < N)
     db_out[bn0 + tid] = dbs[tid];
+}
+
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void mwgrad_kernel(WGemm g) {
+  constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  __shared__ float dbs[64];
+  const int zz = (int)blockIdx.z % g.nz;
+  const int slab = (int)blockIdx.z / g.nz;
+  const WProb& p = g.p[zz];
+  const int m_lo = slab * g.m_chunk;
+  const int m_hi = min(g.M, m_lo + g.m_chunk);
+  float* dw_out = p.dw;
+  float* db_out = p.db;
+  if (g.part) {
+    float* base = g.part
+        + ((int64_t)slab * g.nz + zz) * ((int64_t)g.N * g.K + g.N);
+    dw_out = base;
+    db_out = p.db ? base + (int64_t)g.N * g.K : nullptr;
+  }
+  wgrad_tile_body<BF16, MASK>(p.dy, p.ymask, p.x, dw_out, db_out,
+                              p.db != nullptr, g.M, g.N, g.K, g.lddy,
+                              g.ldx, m_lo, m_hi, (int)blockIdx.x * TB,
+                              (int)blockIdx.y * TB, smem, dbs);
+}
+
+// ---------------------------------------------------------------------------
+// Heterogeneous multi-problem wgrad: every layer's dW/db of a whole
+// backward phase in ONE launch (per-problem shapes; linear block table).
+// The workload is launch-latency-bound at batch 64 (OPTIMIZATION_LOG.md) —
+// collapsing the per-layer wgrad launches buys back their launch floors.
+// No split-M (phase batches are small); fully deterministic.
+// ---------------------------------------------------------------------------
+
+constexpr int MAXW = 12;
+struct WHProb {
+  const float* dy; const float* ymask; const float* x;
+  float* dw; float* db;
+  int M, N, K, lddy, ldx;
+  int bx;      // tiles along N
+  int blk0;    // first linear block id of this problem
+};
+struct WHArgs { WHProb p[MAXW]; int np; };
+
+template <bool BF16>
+__global__ __launch_bounds__(256)
+void mwgrad_het_kernel(WHArgs a) {
+  constexpr int LBYTES = BF16 ? (64 * LDSB2 * 2) : (64 * LDSF2 * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  __shared__ float dbs[64];
+  const int b = (int)blockIdx.x;
+  int zi = 0;
+#pragma unroll
+  for (int i = 1; i < MAXW; ++i)
+    if (i < a.np && b >= a.p[i].blk0) zi = i;
+  const WHProb& p = a.p[zi];
+  const int local = b - p.blk0;
+  const int bn0 = (local % p.bx) * TB;
+  const int bk0 = (local / p.bx) * TB;
+  if (p.ymask)
+    wgrad_tile_body<BF16, true>(p.dy, p.ymask, p.x, p.dw, p.db,
+                                p.db != nullptr, p.M, p.N, p.K, p.lddy,
+                                p.ldx, 0, p.M, bn0, bk0, smem, dbs);
+  else
+    wgrad_tile_body<BF16, false>(p.dy, p.ymask, p.x, p.dw, p.db,
+                                 p.db != nullptr, p.M, p.N, p.K, p.lddy,
+                                 p.ldx, 0, p.M, bn0, bk0, smem, dbs);
 }
 
 // deterministic combine of split-M wgrad slabs: for each problem z,
@@ -1000,7 +1057,7 @@ void adam_t_kernel(float* __restrict__ p, const float* __restrict__ g,
                    float* __restrict__ m, float* __restrict__ v,
                    const int64_t* __restrict__ step, int64_t n,
                    float lr, float b1, float b2, float eps, float wd,
-                   ATArgs ta) {
+                   float* __restrict__ targ, float rho, ATArgs ta) {
   const float t = (float)step[0];
   const float bc1 = 1.f - powf(b1, t);
   const float bc2 = 1.f - powf(b2, t);
@@ -1014,6 +1071,12 @@ void adam_t_kernel(float* __restrict__ p, const float* __restrict__ g,
     v[i] = vi;
     float pn = p[i] - lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
     p[i] = pn;
+    // fused polyak target tracking of the post-Adam parameters —
+    // replaces the standalone polyak launch (reference semantics: the
+    // target averages the q_opt.step()-updated critic, and nothing
+    // mutates critic params between Adam and update_targets,
+    // sac/algorithm.py:139,278)
+    if (targ) targ[i] = rho * targ[i] + (1.f - rho) * pn;
 #pragma unroll
     for (int L = 0; L < MAX_T; ++L) {
       if (L >= ta.n_layers) break;
@@ -1152,6 +1215,40 @@ void mwgrad(std::vector<torch::Tensor> dys,
   }
 }
 
+void mwgrad_het(std::vector<torch::Tensor> dys,
+                std::vector<c10::optional<torch::Tensor>> ymasks,
+                std::vector<torch::Tensor> xs,
+                std::vector<torch::Tensor> dws,
+                std::vector<torch::Tensor> dbs,
+                std::vector<int64_t> Ms, std::vector<int64_t> Ns,
+                std::vector<int64_t> Ks, std::vector<int64_t> lddys,
+                std::vector<int64_t> ldxs, std::vector<int64_t> x_offs) {
+  const int np = (int)dys.size();
+  TORCH_CHECK(np >= 1 && np <= MAXW, "mwgrad_het: 1..12 problems");
+  WHArgs a{};
+  a.np = np;
+  int blk = 0;
+  for (int i = 0; i < np; ++i) {
+    WHProb& p = a.p[i];
+    p.dy = dys[i].data_ptr<float>();
+    p.ymask = fptr(ymasks[i]);
+    p.x = xs[i].data_ptr<float>() + x_offs[i];
+    p.dw = dws[i].data_ptr<float>();
+    p.db = dbs[i].numel() ? dbs[i].data_ptr<float>() : nullptr;
+    p.M = (int)Ms[i]; p.N = (int)Ns[i]; p.K = (int)Ks[i];
+    p.lddy = (int)lddys[i]; p.ldx = (int)ldxs[i];
+    p.bx = (p.N + TB - 1) / TB;
+    p.blk0 = blk;
+    blk += p.bx * ((p.K + TB - 1) / TB);
+  }
+  if (*g_bf16_flag)
+    hipLaunchKernelGGL((mwgrad_het_kernel<true>), dim3(blk), dim3(256), 0,
+                       stream(), a);
+  else
+    hipLaunchKernelGGL((mwgrad_het_kernel<false>), dim3(blk), dim3(256), 0,
+                       stream(), a);
+}
+
 void transpose_multi(std::vector<torch::Tensor> ws,
                      std::vector<torch::Tensor> wts) {
   TArgs t{};
@@ -1272,7 +1369,8 @@ void piloss2(torch::Tensor q1, torch::Tensor q2, torch::Tensor logp,
 void adam_t(torch::Tensor p, torch::Tensor g, torch::Tensor m,
             torch::Tensor v, torch::Tensor step, double lr, double b1,
             double b2, double eps, double wd,
-            std::vector<int64_t> offsets, std::vector<torch::Tensor> wts) {
+            std::vector<int64_t> offsets, std::vector<torch::Tensor> wts,
+            c10::optional<torch::Tensor> targ, double rho) {
   ATArgs ta{};
   ta.n_layers = (int)offsets.size();
   TORCH_CHECK(ta.n_layers <= MAX_T);
@@ -1283,12 +1381,15 @@ void adam_t(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     ta.N[i] = (int)wts[i].size(1);
   }
   int64_t n = p.numel();
+  if (targ.has_value()) TORCH_CHECK(targ->numel() == n);
   int blocks = (int)std::min<int64_t>((n + 255) / 256, 1024);
   hipLaunchKernelGGL(adam_t_kernel, dim3(blocks), dim3(256), 0, stream(),
                      p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      step.data_ptr<int64_t>(), n, (float)lr, (float)b1,
-                     (float)b2, (float)eps, (float)wd, ta);
+                     (float)b2, (float)eps, (float)wd,
+                     targ.has_value() ? targ->data_ptr<float>() : nullptr,
+                     (float)rho, ta);
 }
 
 torch::Tensor tg_eps(int64_t ctr_val, int64_t seed, int64_t R, int64_t A,
@@ -1447,6 +1548,8 @@ void register_fused(pybind11::module_& m) {
   m.def("mgemm", &fused::mgemm,
         "multi-problem MFMA GEMM (fwd-form, strided, masked, sum2)");
   m.def("mwgrad", &fused::mwgrad, "multi-problem wgrad + fused db");
+  m.def("mwgrad_het", &fused::mwgrad_het,
+        "heterogeneous multi-problem wgrad: one launch per phase");
   m.def("transpose_multi", &fused::transpose_multi);
   m.def("gather2", &fused::gather2);
   m.def("tg_fwd2", &fused::tg_fwd2);

@@ -21,8 +21,10 @@ multi-problem MFMA kernels in ops/csrc/fused.hip:
    device scalars; alpha may live on device and be Adam-updated
    in-graph (learned entropy temperature, BASELINE north star).
 
-The whole schedule is captured into a hipGraph (3 graphs + 2 flat-bucket
-RCCL all-reduces when data-parallel).  Reference semantics preserved:
+The whole schedule is captured into ONE hipGraph — data-parallel runs
+record both flat-bucket RCCL all-reduces inside the graph (RCCL supports
+captured collectives), with a 3-graph + host-issued-collective fallback
+if the communicator refuses capture.  Reference semantics preserved:
 update order, losses and polyak follow sac/algorithm.py:115-162,77-81.
 """
 

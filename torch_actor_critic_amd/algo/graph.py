@@ -72,6 +72,25 @@ class GraphedSACUpdate:
         import os
         split = (self.world > 1
                  or os.environ.get("TAC_AMD_SPLIT_GRAPHS") == "1")
+        # record both RCCL all-reduces INSIDE one graph when possible —
+        # one replay per update, no host round-trips (same design as
+        # engine.FusedSACEngine._capture)
+        if (self.world > 1 and comm.backend_name() == "nccl"
+                and os.environ.get("TAC_AMD_GRAPH_COLL", "1") != "0"):
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._phase_critic()
+                    comm.allreduce_grads_capturable(self.q_opt.fp.flat_grad)
+                    self._phase_policy()
+                    comm.allreduce_grads_capturable(self.pi_opt.fp.flat_grad)
+                    self._phase_finish()
+                self.graph = g
+                self._graphs = None
+                return
+            except Exception as e:  # pragma: no cover - fallback path
+                logger.warning("in-graph collective capture failed (%s); "
+                               "falling back to split graphs", e)
         if not split:
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):

@@ -56,6 +56,22 @@ class GraphedSACUpdate:
         self.loss_q_acc = torch.zeros((), **opts)
         self.loss_pi_acc = torch.zeros((), **opts)
 
+        # graph-update optimizations (ops/functional.py): cached
+        # transposed weights refreshed per phase + direct wgrad into the
+        # flat-grad views — active ONLY for this init's warmup+capture
+        from ..ops import functional as Fo
+        self._wt_cache: dict = {}
+        self._critic_weights = [p for p in critic.parameters()
+                                if p.ndim in (2, 4)]
+        self._actor_weights = [p for p in actor.parameters()
+                               if p.ndim in (2, 4)]
+        Fo.set_graph_opt(self._wt_cache, True)
+        try:
+            self._init_graphs(warmup_iters)
+        finally:
+            Fo.set_graph_opt(None, False)
+
+    def _init_graphs(self, warmup_iters: int):
         # -- warmup on a side stream (per torch.cuda.graph contract) ----
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -126,6 +142,9 @@ class GraphedSACUpdate:
 
     def _phase_policy(self):
         self.q_opt.step()
+        if self._wt_cache:
+            from ..ops import functional as Fo
+            Fo.refresh_wt_cache(self._wt_cache, self._critic_weights)
         sac_mod._freeze(self.critic, True)
         self.pi_opt.zero_grad()
         pi_state = self.batch.next_states if self.sac.reference_pi_loss \
@@ -141,6 +160,8 @@ class GraphedSACUpdate:
     def _phase_finish(self):
         self.pi_opt.step()
         from ..ops import functional as Fo
+        if self._wt_cache:
+            Fo.refresh_wt_cache(self._wt_cache, self._actor_weights)
         Fo.polyak_(self.target_flat, self.q_opt.fp.flat, self.sac.polyak)
 
     def _reduce(self, opt: FlatAdam):

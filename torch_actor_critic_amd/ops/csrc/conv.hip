@@ -561,9 +561,12 @@ torch::Tensor conv2d_dgrad(torch::Tensor dy,
 std::vector<torch::Tensor> conv2d_wgrad_multi(
     std::vector<torch::Tensor> dys,
     std::vector<c10::optional<torch::Tensor>> ymasks,
-    std::vector<torch::Tensor> xs, torch::Tensor w, int64_t s) {
+    std::vector<torch::Tensor> xs, torch::Tensor w, int64_t s,
+    std::vector<torch::Tensor> out) {
   const int nz = (int)dys.size();
   TORCH_CHECK(nz >= 1 && nz <= 2);
+  TORCH_CHECK(out.empty() || (int)out.size() == 2 * nz,
+              "out must be [dw0, db0[, dw1, db1]]");
   auto d = dims_of(xs[0], w, s);
   const int K = d.IC * d.KH * d.KW;
   const int M = d.B * d.OH * d.OW;
@@ -580,8 +583,13 @@ std::vector<torch::Tensor> conv2d_wgrad_multi(
   std::vector<torch::Tensor> parts;
   ConvWP p[2] = {};
   for (int z = 0; z < nz; ++z) {
-    outs.push_back(torch::empty_like(w));
-    outs.push_back(torch::empty({d.OC}, w.options()));
+    if (out.empty()) {
+      outs.push_back(torch::empty_like(w));
+      outs.push_back(torch::empty({d.OC}, w.options()));
+    } else {
+      outs.push_back(out[2 * z]);
+      outs.push_back(out[2 * z + 1]);
+    }
     parts.push_back(torch::empty({split, per}, w.options()));
     p[z] = ConvWP{dys[z].data_ptr<float>(),
                   ymasks[z].has_value() ? ymasks[z]->data_ptr<float>()
@@ -616,7 +624,7 @@ std::vector<torch::Tensor> conv2d_wgrad(torch::Tensor dy,
                                         c10::optional<torch::Tensor> ymask,
                                         torch::Tensor x, torch::Tensor w,
                                         int64_t s) {
-  auto o = conv2d_wgrad_multi({dy}, {ymask}, {x}, w, s);
+  auto o = conv2d_wgrad_multi({dy}, {ymask}, {x}, w, s, {});
   return {o[0], o[1]};
 }
 
@@ -632,5 +640,8 @@ void register_conv(pybind11::module_& m) {
   m.def("conv2d_wgrad", &convk::conv2d_wgrad);
   m.def("conv2d_fwd_multi", &convk::conv2d_fwd_multi);
   m.def("conv2d_dgrad_multi", &convk::conv2d_dgrad_multi);
-  m.def("conv2d_wgrad_multi", &convk::conv2d_wgrad_multi);
+  m.def("conv2d_wgrad_multi", &convk::conv2d_wgrad_multi,
+        pybind11::arg("dys"), pybind11::arg("ymasks"), pybind11::arg("xs"),
+        pybind11::arg("w"), pybind11::arg("s"),
+        pybind11::arg("out") = std::vector<torch::Tensor>{});
 }

@@ -664,22 +664,27 @@ constexpr int MAX_T = 12;
 struct TArgs {
   const float* w[MAX_T];
   float* wt[MAX_T];
-  int N[MAX_T], K[MAX_T];
+  int N[MAX_T], K[MAX_T], BS[MAX_T];
   int n_layers, blocks_per_layer;
 };
 
+// BS=1: plain [N,K] -> [K,N] (dense dgrad weights).  BS=kh*kw: conv
+// [OC,IC,kh,kw] -> [IC, OC*kh*kw] block transpose (the dgrad layout of
+// conv2d_dgrad; replaces an aten permute+contiguous per backward).
 __global__ __launch_bounds__(256)
 void transpose_multi_kernel(TArgs t) {
   const int layer = blockIdx.x / t.blocks_per_layer;
   const int slice = blockIdx.x % t.blocks_per_layer;
   if (layer >= t.n_layers) return;
-  const int N = t.N[layer], K = t.K[layer];
-  const int64_t total = (int64_t)N * K;
+  const int N = t.N[layer], K = t.K[layer], BS = t.BS[layer];
+  const int64_t total = (int64_t)N * K * BS;
   const int64_t stride = (int64_t)t.blocks_per_layer * blockDim.x;
   for (int64_t idx = (int64_t)slice * blockDim.x + threadIdx.x; idx < total;
        idx += stride) {
-    int n = (int)(idx / K), k = (int)(idx % K);
-    t.wt[layer][(int64_t)k * N + n] = t.w[layer][idx];
+    int64_t n = idx / ((int64_t)K * BS);
+    int64_t r = idx % ((int64_t)K * BS);
+    int64_t k = r / BS, e = r % BS;
+    t.wt[layer][(k * N + n) * BS + e] = t.w[layer][idx];
   }
 }
 
@@ -1250,7 +1255,8 @@ void mwgrad_het(std::vector<torch::Tensor> dys,
 }
 
 void transpose_multi(std::vector<torch::Tensor> ws,
-                     std::vector<torch::Tensor> wts) {
+                     std::vector<torch::Tensor> wts,
+                     std::vector<int64_t> blocks) {
   TArgs t{};
   t.n_layers = (int)ws.size();
   TORCH_CHECK(t.n_layers <= MAX_T);
@@ -1259,10 +1265,13 @@ void transpose_multi(std::vector<torch::Tensor> ws,
     t.wt[i] = wts[i].data_ptr<float>();
     t.N[i] = (int)ws[i].size(0);
     t.K[i] = (int)ws[i].size(1);
+    t.BS[i] = blocks.empty() ? 1 : (int)blocks[i];
+    TORCH_CHECK((int64_t)t.N[i] * t.K[i] * t.BS[i] == ws[i].numel());
   }
   int64_t max_elems = 0;
   for (int i = 0; i < t.n_layers; ++i)
-    max_elems = std::max(max_elems, (int64_t)t.N[i] * t.K[i]);
+    max_elems = std::max(max_elems,
+                         (int64_t)t.N[i] * t.K[i] * t.BS[i]);
   t.blocks_per_layer = (int)std::min<int64_t>(
       128, std::max<int64_t>(8, (max_elems + 4095) / 4096));
   hipLaunchKernelGGL(transpose_multi_kernel,
@@ -1550,7 +1559,9 @@ void register_fused(pybind11::module_& m) {
   m.def("mwgrad", &fused::mwgrad, "multi-problem wgrad + fused db");
   m.def("mwgrad_het", &fused::mwgrad_het,
         "heterogeneous multi-problem wgrad: one launch per phase");
-  m.def("transpose_multi", &fused::transpose_multi);
+  m.def("transpose_multi", &fused::transpose_multi,
+        pybind11::arg("ws"), pybind11::arg("wts"),
+        pybind11::arg("blocks") = std::vector<int64_t>{});
   m.def("gather2", &fused::gather2);
   m.def("tg_fwd2", &fused::tg_fwd2);
   m.def("tg_bwd2", &fused::tg_bwd2);

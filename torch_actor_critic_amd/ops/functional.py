@@ -67,6 +67,93 @@ def randn_like_philox(t: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# Graph-update optimizations (activated by GraphedSACUpdate only):
+#  * cached transposed weights — backward dgrads read a cache entry that
+#    the update refreshes ONCE per phase after its Adam step, instead of
+#    transposing per backward call;
+#  * direct wgrad — weight/bias gradients are written straight into the
+#    .grad views of the flat-gradient buffer (each layer receives exactly
+#    one contribution per captured update, so overwrite == accumulate
+#    from zero), eliminating autograd's per-parameter accumulate-add
+#    kernels (~46/update on the visual critic).
+# Both are scoped: only active between set_graph_opt(cache, True) and
+# set_graph_opt(None, False), i.e. during graph warmup + capture.
+# ---------------------------------------------------------------------------
+
+_graph_wt_cache = None   # data_ptr(w) -> (w, wt, block)
+_graph_direct_wgrad = False
+
+
+def set_graph_opt(wt_cache, direct_wgrad: bool):
+    global _graph_wt_cache, _graph_direct_wgrad
+    _graph_wt_cache = wt_cache
+    _graph_direct_wgrad = direct_wgrad
+
+
+def _dense_wt(ext, w):
+    """Transposed dense weight [K,N] for dgrad, cached when active."""
+    c = _graph_wt_cache
+    if c is not None:
+        ent = c.get(w.data_ptr())
+        if ent is not None:
+            return ent[1]
+    wt = torch.empty(w.shape[1], w.shape[0], device=w.device,
+                     dtype=w.dtype)
+    ext.transpose_multi([w], [wt])
+    if c is not None:
+        c[w.data_ptr()] = (w, wt, 1)
+    return wt
+
+
+def _conv_wt(ext, w):
+    """Conv dgrad weight layout [IC, OC*kh*kw], cached when active."""
+    oc, ic, kh, kw = w.shape
+    c = _graph_wt_cache
+    if c is not None:
+        ent = c.get(w.data_ptr())
+        if ent is not None:
+            return ent[1]
+        wt = torch.empty(ic, oc * kh * kw, device=w.device, dtype=w.dtype)
+        ext.transpose_multi([w], [wt], [kh * kw])
+        c[w.data_ptr()] = (w, wt, kh * kw)
+        return wt
+    return w.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw).contiguous()
+
+
+def refresh_wt_cache(cache, weights):
+    """Refresh the cached transposed layouts of `weights` (those already
+    registered) in batched transpose_multi launches — called right after
+    the owning optimizer's Adam step, inside the captured graph."""
+    ext = require_extension()
+    ws, wts, bss = [], [], []
+    for w in weights:
+        ent = cache.get(w.data_ptr())
+        if ent is not None:
+            ws.append(ent[0])
+            wts.append(ent[1])
+            bss.append(ent[2])
+    for i in range(0, len(ws), 12):
+        ext.transpose_multi(ws[i:i + 12], wts[i:i + 12], bss[i:i + 12])
+
+
+def _direct_outs(pairs):
+    """(w.grad, b.grad) buffers for in-place wgrad writes, or None when
+    the direct path is off / a .grad view is missing."""
+    if not _graph_direct_wgrad:
+        return None
+    outs = []
+    for w, b in pairs:
+        # frozen weights (e.g. the critic during the policy phase) must
+        # NOT have their phase-1 gradients clobbered — fall back to the
+        # allocating path whose outputs autograd discards
+        if not w.requires_grad or w.grad is None or \
+                (b is not None and (not b.requires_grad or b.grad is None)):
+            return None
+        outs.append((w.grad, b.grad if b is not None else None))
+    return outs
+
+
+# ---------------------------------------------------------------------------
 # Linear (+ optional ReLU)
 # ---------------------------------------------------------------------------
 
@@ -83,26 +170,36 @@ class _NativeLinear(torch.autograd.Function):
         # pipelined multi-problem GEMM kernel (fused.hip), single problem
         ext.mgemm([x], [w], [b], [y], [None], M, N, K, K, N, relu,
                   [], [], [], 0, 0, 0, [])
-        ctx.save_for_backward(x, w, y)
+        ctx.save_for_backward(x, w, y, b)
         ctx.relu = relu
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, w, y = ctx.saved_tensors
+        x, w, y, b = ctx.saved_tensors
         ext = require_extension()
         dy = dy.contiguous()
         M, K = x.shape
         N = w.shape[0]
         ymask = y if ctx.relu else None
-        dw = torch.empty_like(w)
-        db = torch.empty(N, device=w.device, dtype=w.dtype)
-        # pipelined coalesced wgrad (split-M at large batch) + fused db
-        ext.mwgrad([dy], [ymask], [x], [dw], [db], M, N, K, N, K, 0)
+        dw = db = None
+        if ctx.needs_input_grad[1]:
+            outs = _direct_outs([(w, b)])
+            if outs is None:
+                dw = torch.empty_like(w)
+                db = torch.empty(N, device=w.device, dtype=w.dtype)
+            else:
+                dw, db = outs[0]
+                if db is None:
+                    db = torch.empty(0, device=w.device, dtype=w.dtype)
+            # pipelined coalesced wgrad (split-M at large batch) +
+            # fused db
+            ext.mwgrad([dy], [ymask], [x], [dw], [db], M, N, K, N, K, 0)
+            if outs is not None:
+                dw = db = None
         dx = None
         if ctx.needs_input_grad[0]:
-            wt = torch.empty(K, N, device=w.device, dtype=w.dtype)
-            ext.transpose_multi([w], [wt])
+            wt = _dense_wt(ext, w)
             dx = torch.empty_like(x)
             ext.mgemm([dy], [wt], [None], [dx], [ymask], M, K, N, N, K,
                       False, [], [], [], 0, 0, 0, [])
@@ -140,9 +237,9 @@ class _NativeConv2d(torch.autograd.Function):
                            b.contiguous() if b is not None else None,
                            stride, relu)
         if relu:
-            ctx.save_for_backward(x, w, y)
+            ctx.save_for_backward(x, w, y, b)
         else:
-            ctx.save_for_backward(x, w)
+            ctx.save_for_backward(x, w, b)
         ctx.stride = stride
         ctx.relu = relu
         ctx.has_bias = b is not None
@@ -155,17 +252,25 @@ class _NativeConv2d(torch.autograd.Function):
         if ctx.relu:
             # one elementwise relu-backward beats masked gathers inside
             # the conv kernels (measured)
-            x, w, y = ctx.saved_tensors
+            x, w, y, b = ctx.saved_tensors
             dy = torch.ops.aten.threshold_backward(dy, y, 0)
         else:
-            x, w = ctx.saved_tensors
+            x, w, b = ctx.saved_tensors
         dx = None
         if ctx.needs_input_grad[0]:
-            oc, ic, kh, kw = w.shape
-            wt = w.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw).contiguous()
+            wt = _conv_wt(ext, w)
             dx = ext.conv2d_dgrad(dy, None, wt, x, w, ctx.stride)
-        dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
-        return dx, dw, (db if ctx.has_bias else None), None, None
+        dw = db = None
+        if ctx.needs_input_grad[1]:
+            outs = _direct_outs([(w, b)]) if ctx.has_bias else None
+            if outs is not None:
+                ext.conv2d_wgrad_multi([dy], [None], [x], w, ctx.stride,
+                                       out=[outs[0][0], outs[0][1]])
+            else:
+                dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
+                if not ctx.has_bias:
+                    db = None
+        return dx, dw, db, None, None
 
 
 def conv2d(x, w, b, stride: int, relu: bool = False):
@@ -195,13 +300,14 @@ class _PairedLinear(torch.autograd.Function):
         y2 = torch.empty_like(y1)
         ext.mgemm([x1, x2], [w1, w2], [b1, b2], [y1, y2], [None, None],
                   M, N, K, K, N, relu, [], [], [], 0, 0, 0, [])
-        ctx.save_for_backward(x1, x2, w1, w2, y1, y2)
+        ctx.save_for_backward(x1, x2, w1, w2, y1, y2, b1, b2)
         ctx.relu = relu
+        ctx.shared_x = x1 is x2
         return y1, y2
 
     @staticmethod
     def backward(ctx, dy1, dy2):
-        x1, x2, w1, w2, y1, y2 = ctx.saved_tensors
+        x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
         ext = require_extension()
         dy1 = dy1.contiguous()
         dy2 = dy2.contiguous()
@@ -209,22 +315,40 @@ class _PairedLinear(torch.autograd.Function):
         N = w1.shape[0]
         m1 = y1 if ctx.relu else None
         m2 = y2 if ctx.relu else None
-        dw1 = torch.empty_like(w1)
-        dw2 = torch.empty_like(w2)
-        db1 = torch.empty(N, device=w1.device, dtype=w1.dtype)
-        db2 = torch.empty_like(db1)
-        ext.mwgrad([dy1, dy2], [m1, m2], [x1, x2], [dw1, dw2],
-                   [db1, db2], M, N, K, N, K, 0)
+        dw1 = dw2 = db1 = db2 = None
+        if ctx.needs_input_grad[2]:
+            outs = _direct_outs([(w1, b1), (w2, b2)])
+            if outs is None:
+                dw1 = torch.empty_like(w1)
+                dw2 = torch.empty_like(w2)
+                db1 = torch.empty(N, device=w1.device, dtype=w1.dtype)
+                db2 = torch.empty_like(db1)
+                gws = (dw1, db1, dw2, db2)
+            else:
+                (gw1, gb1), (gw2, gb2) = outs
+                if gb1 is None:
+                    gb1 = torch.empty(0, device=w1.device, dtype=w1.dtype)
+                    gb2 = torch.empty_like(gb1)
+                gws = (gw1, gb1, gw2, gb2)
+            ext.mwgrad([dy1, dy2], [m1, m2], [x1, x2],
+                       [gws[0], gws[2]], [gws[1], gws[3]],
+                       M, N, K, N, K, 0)
         dx1 = dx2 = None
         if ctx.needs_input_grad[0]:
-            wt1 = torch.empty(K, N, device=w1.device, dtype=w1.dtype)
-            wt2 = torch.empty_like(wt1)
-            ext.transpose_multi([w1, w2], [wt1, wt2])
+            wt1 = _dense_wt(ext, w1)
+            wt2 = _dense_wt(ext, w2)
             dx1 = torch.empty_like(x1)
-            dx2 = torch.empty_like(x2)
-            ext.mgemm([dy1, dy2], [wt1, wt2], [None, None], [dx1, dx2],
-                      [m1, m2], M, K, N, N, K, False, [], [], [], 0, 0, 0,
-                      [])
+            if ctx.shared_x:
+                # both critics share the input: dx = dy1@wt1 + dy2@wt2
+                # in ONE sum2 GEMM launch; autograd sees dx2=None so no
+                # accumulate-add kernel runs
+                ext.mgemm([dy1], [wt1], [None], [dx1], [m1], M, K, N, N,
+                          K, False, [dy2], [wt2], [m2], N, 0, 0, [])
+            else:
+                dx2 = torch.empty_like(x2)
+                ext.mgemm([dy1, dy2], [wt1, wt2], [None, None],
+                          [dx1, dx2], [m1, m2], M, K, N, N, K, False,
+                          [], [], [], 0, 0, 0, [])
         return dx1, dx2, dw1, db1, dw2, db2, None
 
 
@@ -248,9 +372,9 @@ class _PairedConv2d(torch.autograd.Function):
             [b1.contiguous() if b1 is not None else None,
              b2.contiguous() if b2 is not None else None], stride, relu)
         if relu:
-            ctx.save_for_backward(x1, x2, w1, w2, y1, y2)
+            ctx.save_for_backward(x1, x2, w1, w2, y1, y2, b1, b2)
         else:
-            ctx.save_for_backward(x1, x2, w1, w2)
+            ctx.save_for_backward(x1, x2, w1, w2, b1, b2)
         ctx.stride = stride
         ctx.relu = relu
         ctx.has_bias = b1 is not None
@@ -262,25 +386,31 @@ class _PairedConv2d(torch.autograd.Function):
         dy1 = dy1.contiguous()
         dy2 = dy2.contiguous()
         if ctx.relu:
-            x1, x2, w1, w2, y1, y2 = ctx.saved_tensors
+            x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
             dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
             dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
         else:
-            x1, x2, w1, w2 = ctx.saved_tensors
+            x1, x2, w1, w2, b1, b2 = ctx.saved_tensors
         dx1 = dx2 = None
         if ctx.needs_input_grad[0]:
-            oc, ic, kh, kw = w1.shape
-            wt1 = w1.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw
-                                                 ).contiguous()
-            wt2 = w2.permute(1, 0, 2, 3).reshape(ic, oc * kh * kw
-                                                 ).contiguous()
+            wt1 = _conv_wt(ext, w1)
+            wt2 = _conv_wt(ext, w2)
             dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [None, None],
                                               [wt1, wt2], x1, w1,
                                               ctx.stride)
-        dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
-            [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
-        if not ctx.has_bias:
-            db1 = db2 = None
+        dw1 = db1 = dw2 = db2 = None
+        if ctx.needs_input_grad[2]:
+            outs = (_direct_outs([(w1, b1), (w2, b2)]) if ctx.has_bias
+                    else None)
+            if outs is not None:
+                ext.conv2d_wgrad_multi(
+                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride,
+                    out=[outs[0][0], outs[0][1], outs[1][0], outs[1][1]])
+            else:
+                dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
+                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
+                if not ctx.has_bias:
+                    db1 = db2 = None
         return dx1, dx2, dw1, db1, dw2, db2, None, None
 
 

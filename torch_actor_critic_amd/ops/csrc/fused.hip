@@ -82,6 +82,12 @@ struct MGemm {
   MProb p[MAXZ];
   int M, N, K, lda, ldy;
   int K2;  // reduction depth of the second operand pair (SUM2)
+  // split-K (skinny-M shapes, e.g. the visual 3136->512 dense layer:
+  // an un-split launch is 8 workgroups on 256 CUs).  blockIdx.z =
+  // slab*nz + z; slabs write raw partials, the combine kernel sums
+  // deterministically and applies bias/ReLU.
+  int nz, k_chunk;
+  float* part;  // [split*nz][M*N] partials, or null (no split)
 };
 
 constexpr int TB = 64;        // block tile (M and N)
@@ -320,7 +326,8 @@ DEVINL void mma_tiles_p(const void* xs_, const void* ws_,
 
 template <bool BF16, bool MASK>
 DEVINL void gemm_pass(const float* x, const float* w, const float* mask,
-                      int M, int N, int K, int lda, void* xs, void* ws,
+                      int M, int N, int K, int lda, int ldw,
+                      void* xs, void* ws,
                       int bm0, int bn0, int lane, int wrow, int wcol,
                       int tid, f32x4 (&acc)[2][2]) {
   constexpr int BK = BF16 ? BKP : BKF2;
@@ -328,13 +335,14 @@ DEVINL void gemm_pass(const float* x, const float* w, const float* mask,
   float va[EL], vb[EL];
   (void)tid;
   load_tile_regs<BF16, MASK>(va, x, mask, bm0, 0, M, K, lda);
-  load_tile_regs<BF16, false>(vb, w, nullptr, bn0, 0, N, K, K);
+  load_tile_regs<BF16, false>(vb, w, nullptr, bn0, 0, N, K, ldw);
   for (int k0 = 0; k0 < K; k0 += BK) {
     write_tile_lds<BF16>(xs, va);
     write_tile_lds<BF16>(ws, vb);
     if (k0 + BK < K) {
       load_tile_regs<BF16, MASK>(va, x, mask, bm0, k0 + BK, M, K, lda);
-      load_tile_regs<BF16, false>(vb, w, nullptr, bn0, k0 + BK, N, K, K);
+      load_tile_regs<BF16, false>(vb, w, nullptr, bn0, k0 + BK, N, K,
+                                  ldw);
     }
     __syncthreads();
     mma_tiles_p<BF16>(xs, ws, acc, lane, wrow, wcol);
@@ -345,7 +353,9 @@ DEVINL void gemm_pass(const float* x, const float* w, const float* mask,
 template <bool BF16, bool MASK, bool RELU, bool SUM2>
 __global__ __launch_bounds__(256)
 void mgemm_kernel(MGemm g) {
-  const MProb& p = g.p[blockIdx.z];
+  const int zz = g.part ? ((int)blockIdx.z % g.nz) : (int)blockIdx.z;
+  const int slab = g.part ? ((int)blockIdx.z / g.nz) : 0;
+  const MProb& p = g.p[zz];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -358,15 +368,24 @@ void mgemm_kernel(MGemm g) {
   void* xs = smem;
   void* ws = smem + LBYTES;
 
+  const int k_lo = slab * g.k_chunk;
+  const int k_len = g.part ? min(g.K - k_lo, g.k_chunk) : g.K;
+
   f32x4 acc[2][2] = {};
-  gemm_pass<BF16, MASK>(p.x, p.w, p.mask, g.M, g.N, g.K, g.lda, xs, ws,
+  gemm_pass<BF16, MASK>(p.x + k_lo, p.w + k_lo,
+                        p.mask ? p.mask + k_lo : nullptr,
+                        g.M, g.N, k_len, g.lda, g.K, xs, ws,
                         bm0, bn0, lane, wrow, wcol, tid, acc);
-  if constexpr (SUM2) {
-    gemm_pass<BF16, MASK>(p.x2, p.w2, p.mask2, g.M, g.N, g.K2, g.K2, xs,
-                          ws, bm0, bn0, lane, wrow, wcol, tid, acc);
+  if constexpr (SUM2) {  // split-K never combines with SUM2 (host gate)
+    gemm_pass<BF16, MASK>(p.x2, p.w2, p.mask2, g.M, g.N, g.K2, g.K2,
+                          g.K2, xs, ws, bm0, bn0, lane, wrow, wcol, tid,
+                          acc);
   }
 
   const int crow = (lane >> 4) * 4, ccol = lane & 15;
+  float* part_out = g.part
+      ? g.part + (int64_t)((int64_t)slab * g.nz + zz) * g.M * g.N
+      : nullptr;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -377,11 +396,36 @@ void mgemm_kernel(MGemm g) {
         int gcol = bn0 + wcol + ni * 16 + ccol;
         if (grow < g.M && gcol < g.N) {
           float v = acc[mi][ni][r];
+          if (part_out) {  // raw partial; epilogue runs in the combine
+            part_out[(int64_t)grow * g.N + gcol] = v;
+            continue;
+          }
           if (p.bias) v += p.bias[gcol];
           if constexpr (RELU) v = fmaxf(v, 0.f);
           p.y[(int64_t)grow * g.ldy + gcol] = v;
         }
       }
+}
+
+// deterministic split-K combine: y = act(sum_slab part + bias)
+__global__ __launch_bounds__(256)
+void mgemm_combine_kernel(MGemm g, int split, bool relu) {
+  const int64_t per = (int64_t)g.M * g.N;
+  const int64_t total = per * g.nz;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int z = (int)(i / per);
+    const int64_t off = i % per;
+    float s = 0.f;
+    for (int sl = 0; sl < split; ++sl)
+      s += g.part[((int64_t)sl * g.nz + z) * per + off];
+    const MProb& p = g.p[z];
+    const int m = (int)(off / g.N), n = (int)(off % g.N);
+    if (p.bias) s += p.bias[n];
+    if (relu) s = fmaxf(s, 0.f);
+    p.y[(int64_t)m * g.ldy + n] = s;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1142,8 +1186,28 @@ void mgemm(std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
       if (g.p[z].mask2) g.p[z].mask2 += x2_off;
     }
   }
-  dim3 grid((M + TB - 1) / TB, (N + TB - 1) / TB, nz);
   const bool bf16 = *g_bf16_flag;
+  // split-K for deep-K, few-tile launches (e.g. visual dense 3136->512
+  // at batch 64: 8 workgroups un-split on 256 CUs)
+  const int BKc = bf16 ? BKP : BKF2;
+  const int gx = (int)((M + TB - 1) / TB), gy = (int)((N + TB - 1) / TB);
+  const int tiles = gx * gy * nz;
+  int split = 1;
+  if (!sum2 && K >= 1024 && tiles < 128) {
+    const int kchunks = (int)((K + BKc - 1) / BKc);
+    split = std::max(1, std::min({kchunks, (256 + tiles - 1) / tiles, 16}));
+  }
+  int k_chunk = (int)(((K + split - 1) / split + BKc - 1) / BKc * BKc);
+  split = (int)((K + k_chunk - 1) / k_chunk);
+  g.nz = nz;
+  g.k_chunk = k_chunk;
+  g.part = nullptr;
+  torch::Tensor part;
+  if (split > 1) {
+    part = torch::empty({(int64_t)split * nz, M * N}, ys[0].options());
+    g.part = part.data_ptr<float>();
+  }
+  dim3 grid(gx, gy, nz * split);
   auto L = [&](auto b, auto m, auto r, auto s) {
     hipLaunchKernelGGL((mgemm_kernel<decltype(b)::value, decltype(m)::value,
                                      decltype(r)::value, decltype(s)::value>),
@@ -1160,6 +1224,12 @@ void mgemm(std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
               else D1(std::false_type{}, std::false_type{}); }
   #undef D1
   #undef D2
+  if (split > 1) {
+    int64_t total = (int64_t)M * N * nz;
+    int blocks = (int)std::min<int64_t>((total + 255) / 256, 512);
+    hipLaunchKernelGGL(mgemm_combine_kernel, dim3(blocks), dim3(256), 0,
+                       stream(), g, split, relu);
+  }
 }
 
 void mwgrad(std::vector<torch::Tensor> dys,

@@ -720,6 +720,64 @@ void replay_gather_kernel(const float* __restrict__ state,
 }
 
 // ---------------------------------------------------------------------------
+// Visual replay gather: ONE kernel draws the Philox index and copies
+// features / frames (with u8 -> f32 dequantization) / actions / rewards
+// / done into the static batch — replaces ~18 aten index/copy/decode
+// launches per update in the captured visual SAC graph.
+// ---------------------------------------------------------------------------
+
+template <bool QUANT>
+__global__ __launch_bounds__(256)
+void visual_gather_kernel(const float* __restrict__ feat,
+                          const void* __restrict__ frames,
+                          const float* __restrict__ nfeat,
+                          const void* __restrict__ nframes,
+                          const float* __restrict__ act,
+                          const float* __restrict__ rew,
+                          const float* __restrict__ done,
+                          const int64_t* __restrict__ size_dev,
+                          const int64_t* __restrict__ ctr, uint64_t seed,
+                          float* __restrict__ of, float* __restrict__ ofr,
+                          float* __restrict__ onf,
+                          float* __restrict__ onfr,
+                          float* __restrict__ oa,
+                          float* __restrict__ orew,
+                          float* __restrict__ od,
+                          int feat_dim, int64_t frame_n, int act_dim) {
+  const int j = blockIdx.x;
+  const uint64_t size = (uint64_t)size_dev[0];
+  Philox4 r = philox4(seed, (uint64_t)ctr[0], (uint64_t)j);
+  uint64_t u = ((uint64_t)r.x << 32) | r.y;
+  const int64_t idx = (int64_t)(u % (size ? size : 1));
+
+  for (int c = threadIdx.x; c < feat_dim; c += blockDim.x) {
+    of[(int64_t)j * feat_dim + c] = feat[idx * feat_dim + c];
+    onf[(int64_t)j * feat_dim + c] = nfeat[idx * feat_dim + c];
+  }
+  for (int c = threadIdx.x; c < act_dim; c += blockDim.x)
+    oa[(int64_t)j * act_dim + c] = act[idx * act_dim + c];
+  if constexpr (QUANT) {
+    const uint8_t* f0 = (const uint8_t*)frames + idx * frame_n;
+    const uint8_t* f1 = (const uint8_t*)nframes + idx * frame_n;
+    for (int64_t c = threadIdx.x; c < frame_n; c += blockDim.x) {
+      ofr[(int64_t)j * frame_n + c] = (float)f0[c] / 127.5f - 1.0f;
+      onfr[(int64_t)j * frame_n + c] = (float)f1[c] / 127.5f - 1.0f;
+    }
+  } else {
+    const float* f0 = (const float*)frames + idx * frame_n;
+    const float* f1 = (const float*)nframes + idx * frame_n;
+    for (int64_t c = threadIdx.x; c < frame_n; c += blockDim.x) {
+      ofr[(int64_t)j * frame_n + c] = f0[c];
+      onfr[(int64_t)j * frame_n + c] = f1[c];
+    }
+  }
+  if (threadIdx.x == 0) {
+    orew[j] = rew[idx];
+    od[j] = done[idx];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Philox standard-normal sampler (Box-Muller) — graph-replay-safe noise
 // for the reparameterized policy sample (torch's RNG offset is host-side;
 // this keeps the whole SAC update replayable with fresh noise).
@@ -989,6 +1047,50 @@ void replay_sample_into(torch::Tensor state, torch::Tensor act,
                      od.data_ptr<float>(), obs_dim, act_dim);
 }
 
+void visual_sample_into(torch::Tensor feat, torch::Tensor frames,
+                        torch::Tensor nfeat, torch::Tensor nframes,
+                        torch::Tensor act, torch::Tensor rew,
+                        torch::Tensor done, torch::Tensor size_dev,
+                        torch::Tensor ctr, int64_t seed,
+                        torch::Tensor of, torch::Tensor ofr,
+                        torch::Tensor onf, torch::Tensor onfr,
+                        torch::Tensor oa, torch::Tensor orew,
+                        torch::Tensor od) {
+  CHECK_IN(feat); CHECK_IN(of); CHECK_IN(ofr);
+  const int B = (int)of.size(0);
+  const int feat_dim = (int)feat.size(1);
+  const int act_dim = (int)act.size(1);
+  const int64_t frame_n = frames.numel() / frames.size(0);
+  const bool quant = frames.scalar_type() == torch::kUInt8;
+  auto s = cur_stream();
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, s,
+                     ctr.data_ptr<int64_t>());
+  if (quant)
+    hipLaunchKernelGGL((visual_gather_kernel<true>), dim3(B), dim3(256),
+                       0, s, feat.data_ptr<float>(), frames.data_ptr(),
+                       nfeat.data_ptr<float>(), nframes.data_ptr(),
+                       act.data_ptr<float>(), rew.data_ptr<float>(),
+                       done.data_ptr<float>(),
+                       size_dev.data_ptr<int64_t>(),
+                       ctr.data_ptr<int64_t>(), (uint64_t)seed,
+                       of.data_ptr<float>(), ofr.data_ptr<float>(),
+                       onf.data_ptr<float>(), onfr.data_ptr<float>(),
+                       oa.data_ptr<float>(), orew.data_ptr<float>(),
+                       od.data_ptr<float>(), feat_dim, frame_n, act_dim);
+  else
+    hipLaunchKernelGGL((visual_gather_kernel<false>), dim3(B), dim3(256),
+                       0, s, feat.data_ptr<float>(), frames.data_ptr(),
+                       nfeat.data_ptr<float>(), nframes.data_ptr(),
+                       act.data_ptr<float>(), rew.data_ptr<float>(),
+                       done.data_ptr<float>(),
+                       size_dev.data_ptr<int64_t>(),
+                       ctr.data_ptr<int64_t>(), (uint64_t)seed,
+                       of.data_ptr<float>(), ofr.data_ptr<float>(),
+                       onf.data_ptr<float>(), onfr.data_ptr<float>(),
+                       oa.data_ptr<float>(), orew.data_ptr<float>(),
+                       od.data_ptr<float>(), feat_dim, frame_n, act_dim);
+}
+
 void bump_counter(torch::Tensor ctr) {
   hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, cur_stream(),
                      ctr.data_ptr<int64_t>());
@@ -1054,6 +1156,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step_", &adam_step_);
   m.def("replay_sample", &replay_sample);
   m.def("replay_sample_into", &replay_sample_into);
+  m.def("visual_sample_into", &visual_sample_into);
   m.def("philox_randn_", &philox_randn_);
   m.def("bump_counter", &bump_counter);
 }

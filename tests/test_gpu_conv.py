@@ -288,3 +288,53 @@ def test_visual_fused_sample_gpu():
     buf.sample_into(out)
     torch.cuda.synchronize()
     assert not torch.equal(prev, out.states.features)
+
+
+def test_quad_forward_with_target_matches_sequential():
+    """The 4-problem (target twins + live twins) critic-phase forward
+    equals running target and live DoubleCritics separately, and its
+    backward produces the same live-critic gradients."""
+    import copy
+    import torch
+    from torch_actor_critic_amd.models.visual import VisualDoubleCritic
+    from torch_actor_critic_amd.envs.visual import MultiObservation
+    from torch_actor_critic_amd.ops import functional as Fo
+
+    Fo.set_compute_dtype("fp32")
+    torch.manual_seed(21)
+    B, feat, act_dim, vis = 16, 12, 4, (3, 36, 36)
+    dc = VisualDoubleCritic(feat, act_dim, vis, [32, 32],
+                            [8, 16, 16], [4, 3, 3], [2, 2, 1]).to(DEV)
+    tg = copy.deepcopy(dc)
+    for p in tg.parameters():
+        p.requires_grad_(False)
+
+    s = MultiObservation(torch.randn(B, feat, device=DEV),
+                         torch.rand(B, *vis, device=DEV))
+    ns = MultiObservation(torch.randn(B, feat, device=DEV),
+                          torch.rand(B, *vis, device=DEV))
+    a = torch.rand(B, act_dim, device=DEV) * 2 - 1
+    a2 = torch.rand(B, act_dim, device=DEV) * 2 - 1
+
+    # sequential reference
+    dc.zero_grad()
+    with torch.no_grad():
+        r1t, r2t = tg(ns, a2)
+    r1, r2 = dc(s, a)
+    (r1.sum() + r2.sum()).backward()
+    ref_grads = [p.grad.clone() for p in dc.parameters()
+                 if p.grad is not None]
+
+    dc.zero_grad()
+    q1t, q2t, q1, q2 = dc.forward_with_target(tg, s, a, ns, a2)
+    assert not q1t.requires_grad and not q2t.requires_grad
+    assert torch.allclose(q1t, r1t, atol=1e-4), (q1t - r1t).abs().max()
+    assert torch.allclose(q2t, r2t, atol=1e-4)
+    assert torch.allclose(q1, r1, atol=1e-4)
+    assert torch.allclose(q2, r2, atol=1e-4)
+    (q1.sum() + q2.sum()).backward()
+    got = [p.grad.clone() for p in dc.parameters() if p.grad is not None]
+    assert len(got) == len(ref_grads)
+    for g, r in zip(got, ref_grads):
+        assert torch.allclose(g, r, atol=1e-3, rtol=1e-3), \
+            (g - r).abs().max()

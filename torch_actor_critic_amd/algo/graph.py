@@ -132,13 +132,34 @@ class GraphedSACUpdate:
     def _phase_critic(self):
         self.buffer.sample_into(self.batch)
         self.q_opt.zero_grad()
-        loss_q = sac_mod.eval_q_loss(
-            self.actor, self.critic, self.target_critic,
-            self.batch.states, self.batch.actions, self.batch.rewards,
-            self.batch.next_states, self.batch.done,
-            self.sac.alpha, self.sac.gamma, self.sac.reward_scale)
+        b = self.batch
+        quad = getattr(self.critic, "forward_with_target", None)
+        if quad is not None and self._use_quad():
+            # visual fast path: target twins + live twins share every
+            # conv/GEMM launch (4 problems per launch)
+            from ..ops import functional as Fo
+            with torch.no_grad():
+                a2, logp_ac = self.actor(b.next_states)
+            q1_t, q2_t, q1, q2 = quad(self.target_critic, b.states,
+                                      b.actions, b.next_states, a2)
+            loss_q = Fo.sac_q_loss(q1, q2, q1_t, q2_t, logp_ac,
+                                   b.rewards, b.done, self.sac.alpha,
+                                   self.sac.gamma, self.sac.reward_scale)
+        else:
+            loss_q = sac_mod.eval_q_loss(
+                self.actor, self.critic, self.target_critic,
+                b.states, b.actions, b.rewards,
+                b.next_states, b.done,
+                self.sac.alpha, self.sac.gamma, self.sac.reward_scale)
         loss_q.backward()
         self.loss_q_acc += loss_q.detach()
+
+    def _use_quad(self) -> bool:
+        from ..ops import use_native
+        try:
+            return use_native(self.critic.q1.layers[0].weight)
+        except (AttributeError, IndexError):
+            return False
 
     def _phase_policy(self):
         self.q_opt.step()

@@ -201,6 +201,65 @@ class VisualDoubleCritic(nn.Module):
             return self.q1(state, action), self.q2(state, action)
         return self._forward_paired(state, action)
 
+    def forward_with_target(self, target, state, action,
+                            next_state, next_action):
+        """Critic-phase fast path: run the TARGET twins on
+        (next_state, next_action) and the live twins on (state, action)
+        through 4-problem conv/GEMM launches (one launch per layer for
+        all four streams — the MLP engine's 4-problem trick,
+        algo/engine.py).  Returns (q1_target, q2_target, q1, q2); the
+        target outputs are non-differentiable."""
+        q1, q2 = self.q1, self.q2
+        t1, t2 = target.q1, target.q2
+        img = state.frame
+        img_t = next_state.frame
+        if img.ndim == 3:
+            img = img.view((-1, *q1.vis_dim))
+            img_t = img_t.view((-1, *q1.vis_dim))
+        x = torch.cat([state.features, action], dim=-1)
+        xt = torch.cat([next_state.features, next_action], dim=-1)
+        if x.ndim == 1:
+            x = x.view(-1, q1.obs_dim + q1.act_dim)
+            xt = xt.view(-1, q1.obs_dim + q1.act_dim)
+
+        v1, v2 = q1.visual_network, q2.visual_network
+        tv1, tv2 = t1.visual_network, t2.visual_network
+        ht1 = ht2 = img_t
+        h1 = h2 = img
+        for name in ("conv_0", "conv_1", "conv_2"):
+            c1, c2 = getattr(v1, name), getattr(v2, name)
+            tc1, tc2 = getattr(tv1, name), getattr(tv2, name)
+            ht1, ht2, h1, h2 = Fo.conv2d_quad(
+                ht1, ht2, h1, h2, tc1, tc2, c1, c2, c1.stride[0], True)
+        B = h1.shape[0]
+        ht1, ht2 = ht1.reshape(B, -1), ht2.reshape(B, -1)
+        h1, h2 = h1.reshape(B, -1), h2.reshape(B, -1)
+        ht1, ht2, h1, h2 = Fo.linear_quad(ht1, ht2, h1, h2,
+                                          tv1.linear, tv2.linear,
+                                          v1.linear, v2.linear, False)
+        ct1, ct2, c1out, c2out = Fo.linear_quad(ht1, ht2, h1, h2,
+                                                tv1.final, tv2.final,
+                                                v1.final, v2.final, False)
+
+        # MLP trunks — ReLU on ALL layers incl. the final width-1 layer
+        # (reference parity, SURVEY.md Q6)
+        mt1 = mt2 = xt
+        m1 = m2 = x
+        for lt1, lt2, l1, l2 in zip(t1.layers, t2.layers,
+                                    q1.layers, q2.layers):
+            mt1, mt2, m1, m2 = Fo.linear_quad(mt1, mt2, m1, m2,
+                                              lt1, lt2, l1, l2, True)
+
+        yt1 = torch.cat([mt1, ct1], dim=1)
+        yt2 = torch.cat([mt2, ct2], dim=1)
+        y1 = torch.cat([m1, c1out], dim=1)
+        y2 = torch.cat([m2, c2out], dim=1)
+        ot1, ot2, o1, o2 = Fo.linear_quad(yt1, yt2, y1, y2,
+                                          t1.final, t2.final,
+                                          q1.final, q2.final, False)
+        return (torch.squeeze(ot1, -1), torch.squeeze(ot2, -1),
+                torch.squeeze(o1, -1), torch.squeeze(o2, -1))
+
     def _forward_paired(self, state: MultiObservation, action):
         """Both critics in lockstep: every identically-shaped layer pair
         (convs, dense heads, MLP trunk) runs as ONE multi-problem kernel

@@ -129,10 +129,15 @@ struct ConvP {
   const float* x; const float* w; const float* bias; float* y;
 };
 
+// up to 4 problems per launch (blockIdx.z): the twin critics AND the
+// target twins share every conv launch in the critic phase (same shapes,
+// different inputs/weights) — mirrors the MLP engine's 4-problem GEMMs
+struct ConvQ { ConvP p[4]; };
+
 template <bool BF16, bool RELU>
 __global__ __launch_bounds__(256)
-void conv_fwd_kernel(ConvP p0, ConvP p1, ConvDims d) {
-  const ConvP& pp = blockIdx.z ? p1 : p0;
+void conv_fwd_kernel(ConvQ q, ConvDims d) {
+  const ConvP& pp = q.p[blockIdx.z];
   const float* x = pp.x;
   const float* w = pp.w;
   const float* bias = pp.bias;
@@ -486,16 +491,16 @@ std::vector<torch::Tensor> conv2d_fwd_multi(
     std::vector<c10::optional<torch::Tensor>> biases, int64_t s,
     bool relu) {
   const int nz = (int)xs.size();
-  TORCH_CHECK(nz >= 1 && nz <= 2);
+  TORCH_CHECK(nz >= 1 && nz <= 4);
   auto d = dims_of(xs[0], ws[0], s);
   std::vector<torch::Tensor> ys;
-  ConvP p[2] = {};
+  ConvQ q{};
   for (int z = 0; z < nz; ++z) {
     ys.push_back(torch::empty({d.B, d.OC, d.OH, d.OW}, xs[z].options()));
-    p[z] = ConvP{xs[z].data_ptr<float>(), ws[z].data_ptr<float>(),
-                 biases[z].has_value() ? biases[z]->data_ptr<float>()
-                                       : nullptr,
-                 ys[z].data_ptr<float>()};
+    q.p[z] = ConvP{xs[z].data_ptr<float>(), ws[z].data_ptr<float>(),
+                   biases[z].has_value() ? biases[z]->data_ptr<float>()
+                                         : nullptr,
+                   ys[z].data_ptr<float>()};
   }
   const int M = d.B * d.OH * d.OW;
   dim3 grid((M + TB - 1) / TB, (d.OC + TB - 1) / TB, nz);
@@ -503,7 +508,7 @@ std::vector<torch::Tensor> conv2d_fwd_multi(
   auto L = [&](auto b16, auto rl) {
     hipLaunchKernelGGL((conv_fwd_kernel<decltype(b16)::value,
                                         decltype(rl)::value>),
-                       grid, dim3(256), 0, stream(), p[0], p[1], d);
+                       grid, dim3(256), 0, stream(), q, d);
   };
   if (bf16) { if (relu) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }

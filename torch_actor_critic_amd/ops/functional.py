@@ -426,6 +426,144 @@ def conv2d_pair(x1, x2, w1, b1, w2, b2, stride: int, relu: bool):
 
 
 # ---------------------------------------------------------------------------
+# Quad ops: target-twin AND live-twin critic layers in ONE 4-problem
+# launch (the critic phase runs both stacks on identical shapes — same
+# trick as the MLP engine's 4-problem GEMMs, algo/engine.py).  Problems
+# [0,1] are the target critics: non-differentiable, no backward work.
+# ---------------------------------------------------------------------------
+
+class _QuadLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xt1, xt2, x1, x2, wt1, bt1, wt2, bt2,
+                w1, b1, w2, b2, relu):
+        ext = require_extension()
+        xt1, xt2 = xt1.contiguous(), xt2.contiguous()
+        x1c, x2c = x1.contiguous(), x2.contiguous()
+        M, K = x1c.shape
+        N = w1.shape[0]
+        ys = [torch.empty(M, N, device=x1c.device, dtype=x1c.dtype)
+              for _ in range(4)]
+        ext.mgemm([xt1, xt2, x1c, x2c], [wt1, wt2, w1, w2],
+                  [bt1, bt2, b1, b2], ys, [None] * 4,
+                  M, N, K, K, N, relu, [], [], [], 0, 0, 0, [])
+        ctx.save_for_backward(x1c, x2c, w1, w2, ys[2], ys[3], b1, b2)
+        ctx.relu = relu
+        ctx.shared_x = x1c is x2c
+        ctx.mark_non_differentiable(ys[0], ys[1])
+        return ys[0], ys[1], ys[2], ys[3]
+
+    @staticmethod
+    def backward(ctx, dyt1, dyt2, dy1, dy2):
+        x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
+        ext = require_extension()
+        dy1 = dy1.contiguous()
+        dy2 = dy2.contiguous()
+        M, K = x1.shape
+        N = w1.shape[0]
+        m1 = y1 if ctx.relu else None
+        m2 = y2 if ctx.relu else None
+        dw1 = dw2 = db1 = db2 = None
+        if ctx.needs_input_grad[8]:
+            outs = _direct_outs([(w1, b1), (w2, b2)])
+            if outs is None:
+                dw1 = torch.empty_like(w1)
+                dw2 = torch.empty_like(w2)
+                db1 = torch.empty(N, device=w1.device, dtype=w1.dtype)
+                db2 = torch.empty_like(db1)
+                gws = (dw1, db1, dw2, db2)
+            else:
+                (gw1, gb1), (gw2, gb2) = outs
+                gws = (gw1, gb1, gw2, gb2)
+            ext.mwgrad([dy1, dy2], [m1, m2], [x1, x2],
+                       [gws[0], gws[2]], [gws[1], gws[3]],
+                       M, N, K, N, K, 0)
+        dx1 = dx2 = None
+        if ctx.needs_input_grad[2]:
+            wt1 = _dense_wt(ext, w1)
+            wt2 = _dense_wt(ext, w2)
+            dx1 = torch.empty_like(x1)
+            if ctx.shared_x:
+                ext.mgemm([dy1], [wt1], [None], [dx1], [m1], M, K, N, N,
+                          K, False, [dy2], [wt2], [m2], N, 0, 0, [])
+            else:
+                dx2 = torch.empty_like(x2)
+                ext.mgemm([dy1, dy2], [wt1, wt2], [None, None],
+                          [dx1, dx2], [m1, m2], M, K, N, N, K, False,
+                          [], [], [], 0, 0, 0, [])
+        return (None, None, dx1, dx2, None, None, None, None,
+                dw1, db1, dw2, db2, None)
+
+
+def linear_quad(xt1, xt2, x1, x2, t1, t2, l1, l2, relu: bool):
+    """(target twin, live twin) linear layers, one 4-problem launch;
+    t*/l* are nn.Linear modules."""
+    return _QuadLinear.apply(xt1, xt2, x1, x2,
+                             t1.weight, t1.bias, t2.weight, t2.bias,
+                             l1.weight, l1.bias, l2.weight, l2.bias,
+                             relu)
+
+
+class _QuadConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xt1, xt2, x1, x2, wt1, bt1, wt2, bt2,
+                w1, b1, w2, b2, stride, relu):
+        ext = require_extension()
+        x1c, x2c = x1.contiguous(), x2.contiguous()
+        ys = ext.conv2d_fwd_multi(
+            [xt1.contiguous(), xt2.contiguous(), x1c, x2c],
+            [wt1, wt2, w1, w2], [bt1, bt2, b1, b2], stride, relu)
+        if relu:
+            ctx.save_for_backward(x1c, x2c, w1, w2, ys[2], ys[3], b1, b2)
+        else:
+            ctx.save_for_backward(x1c, x2c, w1, w2, b1, b2)
+        ctx.stride = stride
+        ctx.relu = relu
+        ctx.mark_non_differentiable(ys[0], ys[1])
+        return ys[0], ys[1], ys[2], ys[3]
+
+    @staticmethod
+    def backward(ctx, dyt1, dyt2, dy1, dy2):
+        ext = require_extension()
+        dy1 = dy1.contiguous()
+        dy2 = dy2.contiguous()
+        if ctx.relu:
+            x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
+            dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
+            dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
+        else:
+            x1, x2, w1, w2, b1, b2 = ctx.saved_tensors
+        dx1 = dx2 = None
+        if ctx.needs_input_grad[2]:
+            wt1 = _conv_wt(ext, w1)
+            wt2 = _conv_wt(ext, w2)
+            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [None, None],
+                                              [wt1, wt2], x1, w1,
+                                              ctx.stride)
+        dw1 = db1 = dw2 = db2 = None
+        if ctx.needs_input_grad[8]:
+            outs = _direct_outs([(w1, b1), (w2, b2)])
+            if outs is not None:
+                ext.conv2d_wgrad_multi(
+                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride,
+                    out=[outs[0][0], outs[0][1], outs[1][0], outs[1][1]])
+            else:
+                dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
+                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
+        return (None, None, dx1, dx2, None, None, None, None,
+                dw1, db1, dw2, db2, None, None)
+
+
+def conv2d_quad(xt1, xt2, x1, x2, t1, t2, c1, c2, stride: int,
+                relu: bool):
+    """(target twin, live twin) conv layers, one 4-problem launch;
+    t*/c* are conv modules."""
+    return _QuadConv2d.apply(xt1, xt2, x1, x2,
+                             t1.weight, t1.bias, t2.weight, t2.bias,
+                             c1.weight, c1.bias, c2.weight, c2.bias,
+                             int(stride), relu)
+
+
+# ---------------------------------------------------------------------------
 # Fused tanh-Gaussian head (sample + squash + log-prob)
 # ---------------------------------------------------------------------------
 

@@ -349,6 +349,128 @@ void conv_dgrad_kernel(ConvGP p0, ConvGP p1, ConvDims d) {
 }
 
 // ---------------------------------------------------------------------------
+// Stride-class dgrad (S>1): pixels with (iy%S, ix%S) == (cy, cx) share
+// the SAME valid tap set (ky ≡ cy, kx ≡ cx mod S), so each class runs a
+// dense GEMM over K_cls = OC·ceil((KH-cy)/S)·ceil((KW-cx)/S) instead of
+// the full K = OC·KH·KW with (S²-1)/S² of the taps predicated to zero
+// (4x less MFMA+gather work at S=2, 16x at S=4).
+// grid: (ceil(Mc_max/TB), ceil(IC/TB), nz·S²); blockIdx.z = cls·nz + z.
+// ---------------------------------------------------------------------------
+
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void conv_dgrad_cls_kernel(ConvGP p0, ConvGP p1, ConvDims d, int nz) {
+  const int zi = (int)blockIdx.z;
+  const int zz = zi % nz;
+  const int cls = zi / nz;
+  const int cy = cls / d.S, cx = cls % d.S;
+  const int IHc = (d.IH - cy + d.S - 1) / d.S;
+  const int IWc = (d.IW - cx + d.S - 1) / d.S;
+  const int KHc = d.KH > cy ? (d.KH - cy + d.S - 1) / d.S : 0;
+  const int KWc = d.KW > cx ? (d.KW - cx + d.S - 1) / d.S : 0;
+  const int Mc = d.B * IHc * IWc;
+  const int Kc = d.OC * KHc * KWc;
+  const int bm0 = blockIdx.x * TB;
+  if (bm0 >= Mc) return;
+  const ConvGP& pp = zz ? p1 : p0;
+  const float* dy = pp.dy;
+  const float* ymask = pp.ymask;
+  const float* wt = pp.wt;
+  float* dx = pp.dx;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bn0 = blockIdx.y * TB;
+  constexpr int BK = BF16 ? BKB : BKF;
+  constexpr int EL = BF16 ? 16 : 4;
+  constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
+  __shared__ __attribute__((aligned(16))) char smem[2 * LBYTES];
+  void* xs = smem;
+  void* ws = smem + LBYTES;
+  f32x4 acc[2][2] = {};
+
+  const int row = tid & 63;
+  const int c00 = (tid >> 6) * EL;
+  const int m_my = bm0 + row;
+  int px0, py0, b0;
+  {
+    int m = m_my < Mc ? m_my : 0;
+    px0 = m % IWc;
+    int t1 = m / IWc;
+    py0 = t1 % IHc;
+    b0 = t1 / IHc;
+  }
+  const int n_my = bn0 + row;
+  const int K_full = d.OC * d.KH * d.KW;
+  const float* wtrow = n_my < d.IC ? wt + (int64_t)n_my * K_full : nullptr;
+
+  float va[EL], vb[EL];
+  auto load_chunk = [&](int k0) {
+    int kc = k0 + c00;
+    int kxi = KWc ? kc % KWc : 0;
+    int t2 = KWc ? kc / KWc : 0;
+    int kyi = KHc ? t2 % KHc : 0;
+    int oc = KHc ? t2 / KHc : 0;
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      int k = k0 + c00 + e;
+      float v = 0.f;
+      float wv = 0.f;
+      if (k < Kc) {
+        if (m_my < Mc) {
+          const int oy = py0 - kyi, ox = px0 - kxi;
+          if (oy >= 0 && ox >= 0 && oy < d.OH && ox < d.OW) {
+            int64_t idx = (((int64_t)b0 * d.OC + oc) * d.OH + oy) * d.OW
+                          + ox;
+            v = dy[idx];
+            if constexpr (MASK) v = ymask[idx] > 0.f ? v : 0.f;
+          }
+        }
+        if (wtrow) {
+          const int ky = cy + kyi * d.S, kx = cx + kxi * d.S;
+          wv = wtrow[(int64_t)oc * d.KH * d.KW + ky * d.KW + kx];
+        }
+      }
+      va[e] = v;
+      vb[e] = wv;
+      if (++kxi == KWc) { kxi = 0; if (++kyi == KHc) { kyi = 0; ++oc; } }
+    }
+  };
+  load_chunk(0);
+  for (int k0 = 0; k0 < Kc; k0 += BK) {
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      lds_put<BF16>(xs, row, c00 + e, va[e]);
+      lds_put<BF16>(ws, row, c00 + e, vb[e]);
+    }
+    if (k0 + BK < Kc) load_chunk(k0 + BK);
+    __syncthreads();
+    mma_tiles<BF16>(xs, ws, acc, lane, wrow, wcol);
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = bm0 + wrow + mi * 16 + crow + r;
+        int ic = bn0 + wcol + ni * 16 + ccol;
+        if (m < Mc && ic < d.IC) {
+          const int px = m % IWc, t1 = m / IWc;
+          const int py = t1 % IHc, b = t1 / IHc;
+          const int iy = cy + py * d.S, ix = cx + px * d.S;
+          dx[(((int64_t)b * d.IC + ic) * d.IH + iy) * d.IW + ix] =
+              acc[mi][ni][r];
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
 // conv wgrad: dW[oc, k] = sum_m dYeff[m, oc] im2col[m, k]; db fused.
 // Reduction over M; tiles: rows=oc (64), cols=k (64), i-chunks of BK.
 // ---------------------------------------------------------------------------
@@ -541,9 +663,28 @@ std::vector<torch::Tensor> conv2d_dgrad_multi(
                                         : nullptr,
                   wts[z].data_ptr<float>(), dxs[z].data_ptr<float>()};
   }
+  const bool bf16 = *g_bf16_flag2;
+  if (d.S > 1) {
+    // stride-class decomposition: dense taps per class (no predicated
+    // zeros), S^2 classes fan out over blockIdx.z
+    const int IHc = (d.IH + d.S - 1) / d.S;   // class (0,0) is largest
+    const int IWc = (d.IW + d.S - 1) / d.S;
+    const int Mc = d.B * IHc * IWc;
+    dim3 grid((Mc + TB - 1) / TB, (d.IC + TB - 1) / TB,
+              nz * d.S * d.S);
+    auto L = [&](auto b16, auto mk) {
+      hipLaunchKernelGGL((conv_dgrad_cls_kernel<decltype(b16)::value,
+                                                decltype(mk)::value>),
+                         grid, dim3(256), 0, stream(), p[0], p[1], d, nz);
+    };
+    if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
+                else L(std::true_type{}, std::false_type{}); }
+    else      { if (mask) L(std::false_type{}, std::true_type{});
+                else L(std::false_type{}, std::false_type{}); }
+    return dxs;
+  }
   const int M = d.B * d.IH * d.IW;
   dim3 grid((M + TB - 1) / TB, (d.IC + TB - 1) / TB, nz);
-  const bool bf16 = *g_bf16_flag2;
   auto L = [&](auto b16, auto mk) {
     hipLaunchKernelGGL((conv_dgrad_kernel<decltype(b16)::value,
                                           decltype(mk)::value>),

@@ -744,36 +744,46 @@ void visual_gather_kernel(const float* __restrict__ feat,
                           float* __restrict__ orew,
                           float* __restrict__ od,
                           int feat_dim, int64_t frame_n, int act_dim) {
+  // grid (B, SLICES): every block re-derives the Philox index for its
+  // sample j; blockIdx.y partitions the frame copy so the whole gather
+  // fills the chip instead of B workgroups
   const int j = blockIdx.x;
+  const int slice = blockIdx.y;
+  const int nslice = gridDim.y;
   const uint64_t size = (uint64_t)size_dev[0];
   Philox4 r = philox4(seed, (uint64_t)ctr[0], (uint64_t)j);
   uint64_t u = ((uint64_t)r.x << 32) | r.y;
   const int64_t idx = (int64_t)(u % (size ? size : 1));
 
-  for (int c = threadIdx.x; c < feat_dim; c += blockDim.x) {
-    of[(int64_t)j * feat_dim + c] = feat[idx * feat_dim + c];
-    onf[(int64_t)j * feat_dim + c] = nfeat[idx * feat_dim + c];
+  if (slice == 0) {
+    for (int c = threadIdx.x; c < feat_dim; c += blockDim.x) {
+      of[(int64_t)j * feat_dim + c] = feat[idx * feat_dim + c];
+      onf[(int64_t)j * feat_dim + c] = nfeat[idx * feat_dim + c];
+    }
+    for (int c = threadIdx.x; c < act_dim; c += blockDim.x)
+      oa[(int64_t)j * act_dim + c] = act[idx * act_dim + c];
+    if (threadIdx.x == 0) {
+      orew[j] = rew[idx];
+      od[j] = done[idx];
+    }
   }
-  for (int c = threadIdx.x; c < act_dim; c += blockDim.x)
-    oa[(int64_t)j * act_dim + c] = act[idx * act_dim + c];
+  const int64_t chunk = (frame_n + nslice - 1) / nslice;
+  const int64_t c_lo = (int64_t)slice * chunk;
+  const int64_t c_hi = min(frame_n, c_lo + chunk);
   if constexpr (QUANT) {
     const uint8_t* f0 = (const uint8_t*)frames + idx * frame_n;
     const uint8_t* f1 = (const uint8_t*)nframes + idx * frame_n;
-    for (int64_t c = threadIdx.x; c < frame_n; c += blockDim.x) {
+    for (int64_t c = c_lo + threadIdx.x; c < c_hi; c += blockDim.x) {
       ofr[(int64_t)j * frame_n + c] = (float)f0[c] / 127.5f - 1.0f;
       onfr[(int64_t)j * frame_n + c] = (float)f1[c] / 127.5f - 1.0f;
     }
   } else {
     const float* f0 = (const float*)frames + idx * frame_n;
     const float* f1 = (const float*)nframes + idx * frame_n;
-    for (int64_t c = threadIdx.x; c < frame_n; c += blockDim.x) {
+    for (int64_t c = c_lo + threadIdx.x; c < c_hi; c += blockDim.x) {
       ofr[(int64_t)j * frame_n + c] = f0[c];
       onfr[(int64_t)j * frame_n + c] = f1[c];
     }
-  }
-  if (threadIdx.x == 0) {
-    orew[j] = rew[idx];
-    od[j] = done[idx];
   }
 }
 
@@ -1065,8 +1075,10 @@ void visual_sample_into(torch::Tensor feat, torch::Tensor frames,
   auto s = cur_stream();
   hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, s,
                      ctr.data_ptr<int64_t>());
+  const int slices = (int)std::min<int64_t>(8, (frame_n + 4095) / 4096);
+  dim3 grid(B, std::max(1, slices));
   if (quant)
-    hipLaunchKernelGGL((visual_gather_kernel<true>), dim3(B), dim3(256),
+    hipLaunchKernelGGL((visual_gather_kernel<true>), grid, dim3(256),
                        0, s, feat.data_ptr<float>(), frames.data_ptr(),
                        nfeat.data_ptr<float>(), nframes.data_ptr(),
                        act.data_ptr<float>(), rew.data_ptr<float>(),
@@ -1078,7 +1090,7 @@ void visual_sample_into(torch::Tensor feat, torch::Tensor frames,
                        oa.data_ptr<float>(), orew.data_ptr<float>(),
                        od.data_ptr<float>(), feat_dim, frame_n, act_dim);
   else
-    hipLaunchKernelGGL((visual_gather_kernel<false>), dim3(B), dim3(256),
+    hipLaunchKernelGGL((visual_gather_kernel<false>), grid, dim3(256),
                        0, s, feat.data_ptr<float>(), frames.data_ptr(),
                        nfeat.data_ptr<float>(), nframes.data_ptr(),
                        act.data_ptr<float>(), rew.data_ptr<float>(),

@@ -134,12 +134,25 @@ class GraphedSACUpdate:
         self.q_opt.zero_grad()
         b = self.batch
         quad = getattr(self.critic, "forward_with_target", None)
+        self._pi = self._logp_pi = None
         if quad is not None and self._use_quad():
             # visual fast path: target twins + live twins share every
-            # conv/GEMM launch (4 problems per launch)
+            # conv/GEMM launch (4 problems per launch), and ONE stacked
+            # 2B-row actor forward serves both the Bellman action (rows
+            # :B on next_states, detached) and the policy action (rows
+            # B:, differentiable — consumed by _phase_policy).  Rows
+            # draw independent Philox noise, so this equals the
+            # reference's two separate draws (algo/engine.py docstring).
             from ..ops import functional as Fo
-            with torch.no_grad():
-                a2, logp_ac = self.actor(b.next_states)
+            pi_src = b.next_states if self.sac.reference_pi_loss \
+                else b.states
+            B = b.actions.shape[0]
+            stack = self._stack_obs(b.next_states, pi_src)
+            pi_all, logp_all = self.actor(stack)
+            a2 = pi_all[:B].detach()
+            logp_ac = logp_all[:B].detach()
+            self._pi = pi_all[B:]
+            self._logp_pi = logp_all[B:]
             q1_t, q2_t, q1, q2 = quad(self.target_critic, b.states,
                                       b.actions, b.next_states, a2)
             loss_q = Fo.sac_q_loss(q1, q2, q1_t, q2_t, logp_ac,
@@ -153,6 +166,13 @@ class GraphedSACUpdate:
                 self.sac.alpha, self.sac.gamma, self.sac.reward_scale)
         loss_q.backward()
         self.loss_q_acc += loss_q.detach()
+
+    @staticmethod
+    def _stack_obs(a, b):
+        if hasattr(a, "features"):   # MultiObservation
+            return type(a)(torch.cat([a.features, b.features]),
+                           torch.cat([a.frame, b.frame]))
+        return torch.cat([a, b])
 
     def _use_quad(self) -> bool:
         from ..ops import use_native
@@ -168,9 +188,12 @@ class GraphedSACUpdate:
             Fo.refresh_wt_cache(self._wt_cache, self._critic_weights)
         sac_mod._freeze(self.critic, True)
         self.pi_opt.zero_grad()
-        pi_state = self.batch.next_states if self.sac.reference_pi_loss \
-            else self.batch.states
-        pi, logp = self.actor(pi_state)
+        if self._pi is not None:
+            pi, logp = self._pi, self._logp_pi
+        else:
+            pi_state = self.batch.next_states \
+                if self.sac.reference_pi_loss else self.batch.states
+            pi, logp = self.actor(pi_state)
         q1, q2 = self.critic(self.batch.states, pi)
         from ..ops import functional as Fo
         loss_pi = Fo.sac_pi_loss(q1, q2, logp, self.sac.alpha)

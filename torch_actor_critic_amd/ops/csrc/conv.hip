@@ -589,6 +589,30 @@ void wgrad_combine_kernel(const float* __restrict__ part,
   }
 }
 
+// both problems' combines in ONE launch (parts: [z][slab][per])
+__global__ __launch_bounds__(256)
+void wgrad_combine2_kernel(const float* __restrict__ p0,
+                           const float* __restrict__ p1,
+                           float* __restrict__ dw0, float* __restrict__ db0,
+                           float* __restrict__ dw1, float* __restrict__ db1,
+                           int64_t dw_n, int64_t oc_n, int n_slabs) {
+  const int64_t per = dw_n + oc_n;
+  const int64_t total = 2 * per;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int zz = (int)(i / per);
+    const int64_t off = i % per;
+    const float* part = zz ? p1 : p0;
+    float s = 0.f;
+    for (int z = 0; z < n_slabs; ++z) s += part[(int64_t)z * per + off];
+    float* dw = zz ? dw1 : dw0;
+    float* db = zz ? db1 : db0;
+    if (off < dw_n) dw[off] = s;
+    else if (db) db[off - dw_n] = s;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
@@ -755,12 +779,21 @@ std::vector<torch::Tensor> conv2d_wgrad_multi(
   else      { if (mask) L(std::false_type{}, std::true_type{});
               else L(std::false_type{}, std::false_type{}); }
   int64_t dw_n = (int64_t)d.OC * K;
-  int blocks = (int)std::min<int64_t>((per + 255) / 256, 512);
-  for (int z = 0; z < nz; ++z) {
+  int blocks = (int)std::min<int64_t>((per * nz + 255) / 256, 512);
+  if (nz == 2) {
+    hipLaunchKernelGGL(wgrad_combine2_kernel, dim3(blocks), dim3(256), 0,
+                       stream(), parts[0].data_ptr<float>(),
+                       parts[1].data_ptr<float>(),
+                       outs[0].data_ptr<float>(),
+                       outs[1].data_ptr<float>(),
+                       outs[2].data_ptr<float>(),
+                       outs[3].data_ptr<float>(), dw_n,
+                       (int64_t)d.OC, split);
+  } else {
     hipLaunchKernelGGL(wgrad_combine_kernel, dim3(blocks), dim3(256), 0,
-                       stream(), parts[z].data_ptr<float>(),
-                       outs[2 * z].data_ptr<float>(),
-                       outs[2 * z + 1].data_ptr<float>(), dw_n,
+                       stream(), parts[0].data_ptr<float>(),
+                       outs[0].data_ptr<float>(),
+                       outs[1].data_ptr<float>(), dw_n,
                        (int64_t)d.OC, split);
   }
   return outs;

@@ -721,14 +721,41 @@ void transpose_multi_kernel(TArgs t) {
   const int slice = blockIdx.x % t.blocks_per_layer;
   if (layer >= t.n_layers) return;
   const int N = t.N[layer], K = t.K[layer], BS = t.BS[layer];
+  const float* w = t.w[layer];
+  float* wt = t.wt[layer];
+  const int tid = threadIdx.x;
+  if (BS == 1) {
+    // LDS-tiled: both the global read (along k) and the global write
+    // (along n) are coalesced — the naive element loop writes with
+    // stride N (the 512x3136 dense layer was its worst case)
+    __shared__ float tile[64][65];
+    const int tiles_n = (N + 63) >> 6, tiles_k = (K + 63) >> 6;
+    for (int tt = slice; tt < tiles_n * tiles_k;
+         tt += t.blocks_per_layer) {
+      const int n0 = (tt / tiles_k) << 6, k0 = (tt % tiles_k) << 6;
+      for (int e = tid; e < 4096; e += 256) {
+        const int r = e >> 6, c = e & 63;
+        const int n = n0 + r, k = k0 + c;
+        tile[r][c] = (n < N && k < K) ? w[(int64_t)n * K + k] : 0.f;
+      }
+      __syncthreads();
+      for (int e = tid; e < 4096; e += 256) {
+        const int r = e >> 6, c = e & 63;
+        const int k = k0 + r, n = n0 + c;
+        if (k < K && n < N) wt[(int64_t)k * N + n] = tile[c][r];
+      }
+      __syncthreads();
+    }
+    return;
+  }
   const int64_t total = (int64_t)N * K * BS;
   const int64_t stride = (int64_t)t.blocks_per_layer * blockDim.x;
-  for (int64_t idx = (int64_t)slice * blockDim.x + threadIdx.x; idx < total;
+  for (int64_t idx = (int64_t)slice * blockDim.x + tid; idx < total;
        idx += stride) {
     int64_t n = idx / ((int64_t)K * BS);
     int64_t r = idx % ((int64_t)K * BS);
     int64_t k = r / BS, e = r % BS;
-    t.wt[layer][(k * N + n) * BS + e] = t.w[layer][idx];
+    wt[(k * N + n) * BS + e] = w[idx];
   }
 }
 

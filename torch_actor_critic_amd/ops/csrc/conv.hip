@@ -171,42 +171,16 @@ void conv_fwd_kernel(ConvQ q, ConvDims d) {
   float va[EL], vb[EL];
   auto load_chunk = [&](int k0) {
     KDec kd = kdec((k0 + c00) < K ? (k0 + c00) : 0, d);
-    // successive k within a kx-run read successive ix — float4 the runs
-    // (the per-element gather was the kernel's bottleneck: MACs here
-    // are microseconds, the staging is not)
-    if (m_my < M) {
-      int e = 0;
-      while (e < EL) {
-        int k = k0 + c00 + e;
-        if (k >= K) {
-          for (; e < EL; ++e) va[e] = 0.f;
-          break;
-        }
-        int run = EL - e;
-        if (d.KW - kd.kx < run) run = d.KW - kd.kx;
-        if (K - k < run) run = K - k;
-        const int iy = md.oy * d.S + kd.ky;
-        const int ix = md.ox * d.S + kd.kx;
-        const float* src =
-            &x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
-        int q = 0;
-        for (; q + 4 <= run; q += 4) {
-          float4 f;
-          __builtin_memcpy(&f, src + q, 16);
-          va[e + q] = f.x; va[e + q + 1] = f.y;
-          va[e + q + 2] = f.z; va[e + q + 3] = f.w;
-        }
-        for (; q < run; ++q) va[e + q] = src[q];
-        e += run;
-        kd.kx += run;
-        if (kd.kx >= d.KW) {
-          kd.kx = 0;
-          if (++kd.ky == d.KH) { kd.ky = 0; ++kd.ic; }
-        }
-      }
-    } else {
 #pragma unroll
-      for (int e = 0; e < EL; ++e) va[e] = 0.f;
+    for (int e = 0; e < EL; ++e) {
+      int k = k0 + c00 + e;
+      float v = 0.f;
+      if (m_my < M && k < K) {
+        int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
+        v = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
+      }
+      va[e] = v;
+      kinc(kd, d);
     }
     if (wrow_p && ((K & 3) == 0) && k0 + c00 + EL <= K) {
       const float4* src = (const float4*)(wrow_p + k0 + c00);
@@ -546,85 +520,27 @@ void conv_wgrad_kernel(ConvWP p0, ConvWP p1, ConvDims d, int m_chunk,
   const KDec kd = kdec(k_my < K ? k_my : 0, d);
   for (int i0 = m_lo; i0 < m_hi; i0 += BK) {
     MDec md = mdec((i0 + c00) < M ? (i0 + c00) : 0, d);
-    // successive m within an output row read successive ox — float4
-    // the dy / ymask / (S==1) x runs; the scalar gather was the
-    // kernel's bottleneck
-    float va[EL], vb[EL];
-    {
-      int e = 0;
-      while (e < EL) {
-        int m = i0 + c00 + e;
-        if (m >= m_hi) {
-          for (; e < EL; ++e) { va[e] = 0.f; vb[e] = 0.f; }
-          break;
-        }
-        int run = EL - e;
-        if (d.OW - md.ox < run) run = d.OW - md.ox;
-        if (m_hi - m < run) run = m_hi - m;
-        // A: dy[(b, oc_my, oy, ox..ox+run)]
-        if (oc_my < d.OC) {
-          const int64_t idx0 = (((int64_t)md.b * d.OC + oc_my) * d.OH
-                                + md.oy) * d.OW + md.ox;
-          const float* srcA = dy + idx0;
-          int q = 0;
-          for (; q + 4 <= run; q += 4) {
-            float4 f;
-            __builtin_memcpy(&f, srcA + q, 16);
-            va[e + q] = f.x; va[e + q + 1] = f.y;
-            va[e + q + 2] = f.z; va[e + q + 3] = f.w;
-          }
-          for (; q < run; ++q) va[e + q] = srcA[q];
-          if constexpr (MASK) {
-            const float* srcM = ymask + idx0;
-            q = 0;
-            for (; q + 4 <= run; q += 4) {
-              float4 f;
-              __builtin_memcpy(&f, srcM + q, 16);
-              if (f.x <= 0.f) va[e + q] = 0.f;
-              if (f.y <= 0.f) va[e + q + 1] = 0.f;
-              if (f.z <= 0.f) va[e + q + 2] = 0.f;
-              if (f.w <= 0.f) va[e + q + 3] = 0.f;
-            }
-            for (; q < run; ++q)
-              if (srcM[q] <= 0.f) va[e + q] = 0.f;
-          }
-        } else {
-          for (int q = 0; q < run; ++q) va[e + q] = 0.f;
-        }
-        // B: x[(b, ic, oy*S+ky, (ox..ox+run)*S + kx)]
-        if (k_my < K) {
-          const int iy = md.oy * d.S + kd.ky;
-          const int ix = md.ox * d.S + kd.kx;
-          const float* srcB =
-              &x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW
-                 + ix];
-          if (d.S == 1) {
-            int q = 0;
-            for (; q + 4 <= run; q += 4) {
-              float4 f;
-              __builtin_memcpy(&f, srcB + q, 16);
-              vb[e + q] = f.x; vb[e + q + 1] = f.y;
-              vb[e + q + 2] = f.z; vb[e + q + 3] = f.w;
-            }
-            for (; q < run; ++q) vb[e + q] = srcB[q];
-          } else {
-            for (int q = 0; q < run; ++q) vb[e + q] = srcB[q * d.S];
-          }
-        } else {
-          for (int q = 0; q < run; ++q) vb[e + q] = 0.f;
-        }
-        e += run;
-        md.ox += run;
-        if (md.ox >= d.OW) {
-          md.ox = 0;
-          if (++md.oy == d.OH) { md.oy = 0; ++md.b; }
-        }
-      }
-    }
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
-      lds_put<BF16>(smem, row, c00 + e, va[e]);
-      lds_put<BF16>(smem + LBYTES, row, c00 + e, vb[e]);
+      int i = c00 + e;
+      int m = i0 + i;
+      // A tile: as[oc][i] = dYeff[m, oc_my]
+      float va = 0.f;
+      if (m < m_hi && oc_my < d.OC) {
+        int64_t idx = (((int64_t)md.b * d.OC + oc_my) * d.OH + md.oy)
+                          * d.OW + md.ox;
+        va = dy[idx];
+        if constexpr (MASK) va = ymask[idx] > 0.f ? va : 0.f;
+      }
+      lds_put<BF16>(smem, row, i, va);
+      // B tile: bs[k][i] = im2col[m, k_my]
+      float vb = 0.f;
+      if (m < m_hi && k_my < K) {
+        int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
+        vb = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
+      }
+      lds_put<BF16>(smem + LBYTES, row, i, vb);
+      minc(md, d);
     }
     __syncthreads();
     mma_tiles<BF16>(smem, smem + LBYTES, acc, lane, wrow, wcol);

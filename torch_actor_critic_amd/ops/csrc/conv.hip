@@ -573,6 +573,122 @@ void conv_wgrad_kernel(ConvWP p0, ConvWP p1, ConvDims d, int m_chunk,
     db_p[bn0 + tid] = dbs[tid];
 }
 
+// Shared-input twin wgrad: both critics read the SAME im2col (the twin
+// critics see one image batch), so each block stages the x-tile ONCE
+// and runs both problems' MFMAs against it — the scattered im2col
+// gather is the kernel's dominant cost and it halves.
+// grid: (tiles_oc, tiles_k, split); both problems per block.
+template <bool BF16, bool MASK>
+__global__ __launch_bounds__(256)
+void conv_wgrad2s_kernel(ConvWP p0, ConvWP p1, ConvDims d, int m_chunk,
+                         int split) {
+  const int slab = (int)blockIdx.z;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wrow = (wid >> 1) * 32;
+  const int wcol = (wid & 1) * 32;
+  const int bn0 = blockIdx.x * TB;
+  const int bk0 = blockIdx.y * TB;
+  const int M = d.B * d.OH * d.OW;
+  const int K = d.IC * d.KH * d.KW;
+  const int m_lo = slab * m_chunk;
+  const int m_hi = min(M, m_lo + m_chunk);
+  const int64_t per = (int64_t)d.OC * K + d.OC;
+  float* dw_p0 = p0.part + (int64_t)slab * per;
+  float* db_p0 = dw_p0 + (int64_t)d.OC * K;
+  float* dw_p1 = p1.part + (int64_t)slab * per;
+  float* db_p1 = dw_p1 + (int64_t)d.OC * K;
+  (void)split;
+  constexpr int BK = BF16 ? BKB : BKF;
+  constexpr int EL = BF16 ? 16 : 4;
+  constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
+  __shared__ __attribute__((aligned(16))) char smem[3 * LBYTES];
+  __shared__ float dbs0[64], dbs1[64];
+  if (tid < 64) { dbs0[tid] = 0.f; dbs1[tid] = 0.f; }
+  f32x4 acc0[2][2] = {}, acc1[2][2] = {};
+
+  const int row = tid & 63;
+  const int c00 = (tid >> 6) * EL;
+  const int oc_my = bn0 + row;
+  const int k_my = bk0 + row;
+  const KDec kd = kdec(k_my < K ? k_my : 0, d);
+  for (int i0 = m_lo; i0 < m_hi; i0 += BK) {
+    MDec md = mdec((i0 + c00) < M ? (i0 + c00) : 0, d);
+#pragma unroll
+    for (int e = 0; e < EL; ++e) {
+      int i = c00 + e;
+      int m = i0 + i;
+      float va0 = 0.f, va1 = 0.f, vb = 0.f;
+      if (m < m_hi) {
+        if (oc_my < d.OC) {
+          int64_t idx = (((int64_t)md.b * d.OC + oc_my) * d.OH + md.oy)
+                            * d.OW + md.ox;
+          va0 = p0.dy[idx];
+          va1 = p1.dy[idx];
+          if constexpr (MASK) {
+            va0 = p0.ymask[idx] > 0.f ? va0 : 0.f;
+            va1 = p1.ymask[idx] > 0.f ? va1 : 0.f;
+          }
+        }
+        if (k_my < K) {
+          int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
+          vb = p0.x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW
+                    + ix];
+        }
+      }
+      lds_put<BF16>(smem, row, i, va0);
+      lds_put<BF16>(smem + LBYTES, row, i, va1);
+      lds_put<BF16>(smem + 2 * LBYTES, row, i, vb);
+      minc(md, d);
+    }
+    __syncthreads();
+    mma_tiles<BF16>(smem, smem + 2 * LBYTES, acc0, lane, wrow, wcol);
+    mma_tiles<BF16>(smem + LBYTES, smem + 2 * LBYTES, acc1, lane, wrow,
+                    wcol);
+    if (blockIdx.y == 0 && tid < 64) {
+      float s0 = 0.f, s1 = 0.f;
+      if constexpr (BF16) {
+        const __bf16* a0 = (const __bf16*)smem;
+        const __bf16* a1 = (const __bf16*)(smem + LBYTES);
+        for (int i = 0; i < BKB; ++i) {
+          s0 += (float)a0[tid * LDB + i];
+          s1 += (float)a1[tid * LDB + i];
+        }
+      } else {
+        const float* a0 = (const float*)smem;
+        const float* a1 = (const float*)(smem + LBYTES);
+        for (int i = 0; i < BKF; ++i) {
+          s0 += a0[tid * LDF + i];
+          s1 += a1[tid * LDF + i];
+        }
+      }
+      dbs0[tid] += s0;
+      dbs1[tid] += s1;
+    }
+    __syncthreads();
+  }
+
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int oc = bn0 + wrow + mi * 16 + crow + r;
+        int k = bk0 + wcol + ni * 16 + ccol;
+        if (oc < d.OC && k < K) {
+          dw_p0[(int64_t)oc * K + k] = acc0[mi][ni][r];
+          dw_p1[(int64_t)oc * K + k] = acc1[mi][ni][r];
+        }
+      }
+  if (blockIdx.y == 0 && tid < 64 && bn0 + tid < d.OC) {
+    db_p0[bn0 + tid] = dbs0[tid];
+    db_p1[bn0 + tid] = dbs1[tid];
+  }
+}
+
 // deterministic slab combine: dw[i] = sum_z part[z][i]; db likewise
 __global__ __launch_bounds__(256)
 void wgrad_combine_kernel(const float* __restrict__ part,
@@ -740,7 +856,11 @@ std::vector<torch::Tensor> conv2d_wgrad_multi(
   auto d = dims_of(xs[0], w, s);
   const int K = d.IC * d.KH * d.KW;
   const int M = d.B * d.OH * d.OW;
-  const int tiles = ((d.OC + TB - 1) / TB) * ((K + TB - 1) / TB) * nz;
+  // twin critics share the image batch: one x-tile stage serves both
+  // problems (conv_wgrad2s_kernel)
+  const bool shared = nz == 2 && xs[0].data_ptr() == xs[1].data_ptr();
+  const int tiles = ((d.OC + TB - 1) / TB) * ((K + TB - 1) / TB)
+                    * (shared ? 1 : nz);
   const int BKc = 64;
   int max_split = (M + BKc - 1) / BKc;
   int split = std::max(1, std::min({max_split,
@@ -766,13 +886,20 @@ std::vector<torch::Tensor> conv2d_wgrad_multi(
                                         : nullptr,
                   xs[z].data_ptr<float>(), parts[z].data_ptr<float>()};
   }
-  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB, nz * split);
+  dim3 grid((d.OC + TB - 1) / TB, (K + TB - 1) / TB,
+            (shared ? 1 : nz) * split);
   const bool bf16 = *g_bf16_flag2;
   auto L = [&](auto b16, auto mk) {
-    hipLaunchKernelGGL((conv_wgrad_kernel<decltype(b16)::value,
-                                          decltype(mk)::value>),
-                       grid, dim3(256), 0, stream(), p[0], p[1], d,
-                       m_chunk, split);
+    if (shared)
+      hipLaunchKernelGGL((conv_wgrad2s_kernel<decltype(b16)::value,
+                                              decltype(mk)::value>),
+                         grid, dim3(256), 0, stream(), p[0], p[1], d,
+                         m_chunk, split);
+    else
+      hipLaunchKernelGGL((conv_wgrad_kernel<decltype(b16)::value,
+                                            decltype(mk)::value>),
+                         grid, dim3(256), 0, stream(), p[0], p[1], d,
+                         m_chunk, split);
   };
   if (bf16) { if (mask) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }

@@ -134,7 +134,11 @@ struct ConvP {
 // different inputs/weights) — mirrors the MLP engine's 4-problem GEMMs
 struct ConvQ { ConvP p[4]; };
 
-template <bool BF16, bool RELU>
+// TK/TS: compile-time kernel size / stride for the reference conv
+// family ({8,4}, {4,2}, {3,1}); 0 = runtime fallback.  The gathers are
+// VALU-bound (measured ~85 VALU insts per MFMA), so constant-folding
+// the wrap/index arithmetic and using 32-bit addressing is the lever.
+template <bool BF16, bool RELU, int TK, int TS>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(ConvQ q, ConvDims d) {
   const ConvP& pp = q.p[blockIdx.z];
@@ -142,6 +146,9 @@ void conv_fwd_kernel(ConvQ q, ConvDims d) {
   const float* w = pp.w;
   const float* bias = pp.bias;
   float* y = pp.y;
+  const int KW = TK ? TK : d.KW;
+  const int KH = TK ? TK : d.KH;
+  const int S = TS ? TS : d.S;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
@@ -150,7 +157,7 @@ void conv_fwd_kernel(ConvQ q, ConvDims d) {
   const int bm0 = blockIdx.x * TB;
   const int bn0 = blockIdx.y * TB;
   const int M = d.B * d.OH * d.OW;
-  const int K = d.IC * d.KH * d.KW;
+  const int K = d.IC * KH * KW;
   constexpr int BK = BF16 ? BKB : BKF;
   constexpr int EL = BF16 ? 16 : 4;
   constexpr int LBYTES = BF16 ? (64 * LDB * 2) : (64 * LDF * 4);
@@ -165,22 +172,27 @@ void conv_fwd_kernel(ConvQ q, ConvDims d) {
   const MDec md = mdec(m_my < M ? m_my : 0, d);
   const int n_my = bn0 + row;
   const float* wrow_p = n_my < d.OC ? w + (int64_t)n_my * K : nullptr;
+  // 32-bit base of this thread's pixel (host checks numel < 2^31)
+  const int xbase = (md.b * d.IC * d.IH + md.oy * S) * d.IW + md.ox * S;
+  const int ihw = d.IH * d.IW;
 
   // T14 register-staged pipeline: gather tile t into regs, write LDS,
   // prefetch tile t+1 while the MFMAs run.
   float va[EL], vb[EL];
   auto load_chunk = [&](int k0) {
-    KDec kd = kdec((k0 + c00) < K ? (k0 + c00) : 0, d);
+    int kc = (k0 + c00) < K ? (k0 + c00) : 0;
+    int kx = kc % KW;
+    int t2 = kc / KW;
+    int ky = t2 % KH;
+    int ic = t2 / KH;
 #pragma unroll
     for (int e = 0; e < EL; ++e) {
       int k = k0 + c00 + e;
       float v = 0.f;
-      if (m_my < M && k < K) {
-        int iy = md.oy * d.S + kd.ky, ix = md.ox * d.S + kd.kx;
-        v = x[(((int64_t)md.b * d.IC + kd.ic) * d.IH + iy) * d.IW + ix];
-      }
+      if (m_my < M && k < K)
+        v = x[xbase + ic * ihw + ky * d.IW + kx];
       va[e] = v;
-      kinc(kd, d);
+      if (++kx == KW) { kx = 0; if (++ky == KH) { ky = 0; ++ic; } }
     }
     if (wrow_p && ((K & 3) == 0) && k0 + c00 + EL <= K) {
       const float4* src = (const float4*)(wrow_p + k0 + c00);
@@ -765,12 +777,27 @@ std::vector<torch::Tensor> conv2d_fwd_multi(
                    ys[z].data_ptr<float>()};
   }
   const int M = d.B * d.OH * d.OW;
+  for (int z = 0; z < nz; ++z)
+    TORCH_CHECK(xs[z].numel() < INT32_MAX, "conv fwd: 32-bit indexing");
   dim3 grid((M + TB - 1) / TB, (d.OC + TB - 1) / TB, nz);
   const bool bf16 = *g_bf16_flag2;
   auto L = [&](auto b16, auto rl) {
-    hipLaunchKernelGGL((conv_fwd_kernel<decltype(b16)::value,
-                                        decltype(rl)::value>),
-                       grid, dim3(256), 0, stream(), q, d);
+    auto LS = [&](auto tk, auto ts) {
+      hipLaunchKernelGGL((conv_fwd_kernel<decltype(b16)::value,
+                                          decltype(rl)::value,
+                                          decltype(tk)::value,
+                                          decltype(ts)::value>),
+                         grid, dim3(256), 0, stream(), q, d);
+    };
+    using i0 = std::integral_constant<int, 0>;
+    if (d.KW == d.KH && d.KW == 8 && d.S == 4)
+      LS(std::integral_constant<int, 8>{}, std::integral_constant<int, 4>{});
+    else if (d.KW == d.KH && d.KW == 4 && d.S == 2)
+      LS(std::integral_constant<int, 4>{}, std::integral_constant<int, 2>{});
+    else if (d.KW == d.KH && d.KW == 3 && d.S == 1)
+      LS(std::integral_constant<int, 3>{}, std::integral_constant<int, 1>{});
+    else
+      LS(i0{}, i0{});
   };
   if (bf16) { if (relu) L(std::true_type{}, std::true_type{});
               else L(std::true_type{}, std::false_type{}); }

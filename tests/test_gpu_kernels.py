@@ -379,3 +379,47 @@ def test_mwgrad_split_m_parity(ext, M, N, K, mask):
         (db.cpu() - ref_db).abs().max()
     assert torch.allclose(dw.cpu(), ref_dw, atol=2e-3, rtol=1e-4), \
         (dw.cpu() - ref_dw).abs().max()
+
+
+def test_graph_opt_direct_wgrad_and_wt_cache():
+    """The GraphedSACUpdate fast paths — direct wgrad writes into the
+    .grad views and per-phase cached weight transposes — produce the
+    same gradients as the plain autograd path."""
+    import torch
+    from torch_actor_critic_amd.ops import functional as Fo
+
+    Fo.set_compute_dtype("fp32")
+    torch.manual_seed(31)
+    x = torch.randn(32, 24, device=DEV, requires_grad=True)
+    w = torch.randn(16, 24, device=DEV, requires_grad=True)
+    b = torch.randn(16, device=DEV, requires_grad=True)
+
+    # reference: plain path
+    y = Fo.linear_relu(x, w, b, relu=True)
+    y.square().sum().backward()
+    ref_dx, ref_dw, ref_db = x.grad.clone(), w.grad.clone(), b.grad.clone()
+
+    # graph-opt path: grads land in-place, dgrad reads the cached wt
+    x2 = x.detach().clone().requires_grad_(True)
+    w.grad = torch.zeros_like(w)
+    b.grad = torch.zeros_like(b)
+    cache = {}
+    Fo.set_graph_opt(cache, True)
+    try:
+        y2 = Fo.linear_relu(x2, w, b, relu=True)
+        y2.square().sum().backward()
+    finally:
+        Fo.set_graph_opt(None, False)
+    torch.cuda.synchronize()
+    assert w.data_ptr() in cache
+    assert torch.allclose(x2.grad, ref_dx, atol=1e-4)
+    assert torch.allclose(w.grad, ref_dw, atol=1e-4)
+    assert torch.allclose(b.grad, ref_db, atol=1e-4)
+
+    # after a weight update + refresh, the cached transpose tracks w
+    with torch.no_grad():
+        w.mul_(1.5)
+    Fo.refresh_wt_cache(cache, [w])
+    torch.cuda.synchronize()
+    assert torch.allclose(cache[w.data_ptr()][1], w.t().contiguous(),
+                          atol=1e-6)

@@ -1,3 +1,5 @@
+import os, sys
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
 import torch, numpy as np
 from copy import deepcopy
 from torch_actor_critic_amd.algo.graph import GraphedSACUpdate

@@ -41,6 +41,12 @@ def _worker_allreduce(rank, world, port, q):
     comm.allreduce_grads(fp.flat_grad)
     q.put(("grads", rank, fp.flat_grad.clone().numpy()))
 
+    # 2b) the capture-safe variant (recorded in-graph on RCCL) has the
+    # same semantics
+    fp.flat_grad.fill_(float(rank + 1))
+    comm.allreduce_grads_capturable(fp.flat_grad)
+    q.put(("grads2", rank, fp.flat_grad.clone().numpy()))
+
     # 3) epoch stats gather
     stats = comm.gather_stats([float(rank), float(rank) + 10.0])
     q.put(("stats", rank, np.array(sorted(stats))))
@@ -59,7 +65,7 @@ def test_two_rank_collectives():
     for p in procs:
         p.start()
     results = {}
-    for _ in range(6):
+    for _ in range(8):
         kind, rank, *payload = q.get(timeout=110)
         results[(kind, rank)] = payload
     for p in procs:
@@ -75,6 +81,8 @@ def test_two_rank_collectives():
     # grads: mean of 1 and 2 = 1.5 on both ranks
     np.testing.assert_allclose(results[("grads", 0)][0], 1.5)
     np.testing.assert_allclose(results[("grads", 1)][0], 1.5)
+    np.testing.assert_allclose(results[("grads2", 0)][0], 1.5)
+    np.testing.assert_allclose(results[("grads2", 1)][0], 1.5)
 
     # stats: union of both ranks' lists
     np.testing.assert_allclose(results[("stats", 0)][0],
@@ -177,3 +185,23 @@ def test_dp_sac_train_end_to_end():
     np.testing.assert_allclose(out[0][0], out[1][0], atol=1e-6)
     np.testing.assert_allclose(out[0][1], out[1][1], atol=1e-6)
     assert np.isfinite(out[0][2]) and out[0][2] != 0.0
+
+
+def test_gpu_fork_launches_ranks(tmp_path):
+    """gpu_fork re-launches the script as n ranks with torchrun-style
+    env vars (the reference's mpi_fork UX, fixed per SURVEY.md Q4)."""
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = tmp_path / "forked.py"
+    script.write_text(
+        "import os, sys\n"
+        f"sys.path.insert(0, {repo!r})\n"
+        "from torch_actor_critic_amd.parallel.launch import gpu_fork\n"
+        "gpu_fork(2)\n"
+        "print('rank=' + os.environ['RANK'], 'world='\n"
+        "      + os.environ['WORLD_SIZE'], flush=True)\n")
+    out = subprocess.run([sys.executable, str(script)], check=True,
+                         capture_output=True, text=True, timeout=60)
+    assert "rank=0 world=2" in out.stdout
+    assert "rank=1 world=2" in out.stdout

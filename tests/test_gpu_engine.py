@@ -538,3 +538,31 @@ def test_engine_in_graph_collectives(monkeypatch):
         assert np.isfinite(lq) and np.isfinite(lp) and lq > 0
     finally:
         dist.destroy_process_group()
+
+
+def test_engine_bitwise_deterministic():
+    """Two engines built from identical state with the same Philox seed
+    replay to BITWISE-identical parameters — the whole schedule is
+    atomic-free (split-K/split-M use deterministic combines)."""
+    results = []
+    for _ in range(2):
+        sac, actor, critic, target, buf, pi_opt, q_opt, target_flat, _ = \
+            _setup()
+        rng = np.random.default_rng(17)
+        buf.store_batch(rng.standard_normal((512, O)).astype(np.float32),
+                        rng.standard_normal((512, A)).astype(np.float32),
+                        rng.standard_normal(512).astype(np.float32),
+                        rng.standard_normal((512, O)).astype(np.float32),
+                        np.zeros(512, dtype=np.float32))
+        from torch_actor_critic_amd.algo.engine import FusedSACEngine
+        eng = FusedSACEngine(sac, actor, critic, target, buf, pi_opt,
+                             q_opt, target_flat, B, torch.device(DEV),
+                             sample=True, capture=True, philox_seed=99)
+        for _ in range(5):
+            eng.step()
+        torch.cuda.synchronize()
+        results.append((pi_opt.fp.flat.clone(), q_opt.fp.flat.clone(),
+                        target_flat.clone()))
+    assert torch.equal(results[0][0], results[1][0])
+    assert torch.equal(results[0][1], results[1][1])
+    assert torch.equal(results[0][2], results[1][2])

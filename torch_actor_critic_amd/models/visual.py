@@ -209,6 +209,12 @@ class VisualDoubleCritic(nn.Module):
         all four streams — the MLP engine's 4-problem trick,
         algo/engine.py).  Returns (q1_target, q2_target, q1, q2); the
         target outputs are non-differentiable."""
+        from ..ops import use_native
+        if not use_native(state.features, self.q1.layers[0].weight):
+            with torch.no_grad():
+                qt1, qt2 = target(next_state, next_action)
+            q1, q2 = self(state, action)
+            return qt1, qt2, q1, q2
         q1, q2 = self.q1, self.q2
         t1, t2 = target.q1, target.q2
         img = state.frame

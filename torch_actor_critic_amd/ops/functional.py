@@ -474,6 +474,9 @@ class _QuadLinear(torch.autograd.Function):
                 gws = (dw1, db1, dw2, db2)
             else:
                 (gw1, gb1), (gw2, gb2) = outs
+                if gb1 is None:
+                    gb1 = torch.empty(0, device=w1.device, dtype=w1.dtype)
+                    gb2 = torch.empty_like(gb1)
                 gws = (gw1, gb1, gw2, gb2)
             ext.mwgrad([dy1, dy2], [m1, m2], [x1, x2],
                        [gws[0], gws[2]], [gws[1], gws[3]],

@@ -190,11 +190,11 @@ class FusedSACEngine:
         # TAC_AMD_WGRAD_STREAM=1; default off.
         self._use_side = os.environ.get("TAC_AMD_WGRAD_STREAM") == "1"
         # Batch every wgrad of a backward phase into ONE heterogeneous
-        # multi-problem launch (launch-latency win at small batch; the
-        # split-M path is better at large batch, so gate on B).
-        # TAC_AMD_WGRAD_BATCH=0 disables for A/B.
+        # multi-problem launch (+ one phase-wide combine when split-M
+        # kicks in at large batch).  TAC_AMD_WGRAD_BATCH=0 disables
+        # for A/B.
         self._wgrad_batch = (os.environ.get("TAC_AMD_WGRAD_BATCH", "1")
-                             == "1" and B <= 256 and not self._use_side)
+                             == "1" and not self._use_side)
         self._wjobs: t.List[tuple] = []
         self._s2 = torch.cuda.Stream()
         self._fork_evs = [torch.cuda.Event() for _ in range(12)]

@@ -652,8 +652,17 @@ struct WHProb {
   int M, N, K, lddy, ldx;
   int bx;      // tiles along N
   int blk0;    // first linear block id of this problem
+  int64_t poff;  // element offset of this problem in a partial slab
 };
-struct WHArgs { WHProb p[MAXW]; int np; };
+struct WHArgs {
+  WHProb p[MAXW];
+  int np;
+  // split-M (large batch): blockIdx.z = slab; slabs write raw partials
+  // at part[slab*per_slab + p.poff + ...], one combine for EVERYTHING
+  float* part;
+  int m_chunk;
+  int64_t per_slab;
+};
 
 template <bool BF16>
 __global__ __launch_bounds__(256)
@@ -670,14 +679,46 @@ void mwgrad_het_kernel(WHArgs a) {
   const int local = b - p.blk0;
   const int bn0 = (local % p.bx) * TB;
   const int bk0 = (local / p.bx) * TB;
+  float* dw_out = p.dw;
+  float* db_out = p.db;
+  int m_lo = 0, m_hi = p.M;
+  if (a.part) {
+    float* base = a.part + (int64_t)blockIdx.z * a.per_slab + p.poff;
+    dw_out = base;
+    db_out = p.db ? base + (int64_t)p.N * p.K : nullptr;
+    m_lo = (int)blockIdx.z * a.m_chunk;
+    m_hi = min(p.M, m_lo + a.m_chunk);
+  }
   if (p.ymask)
-    wgrad_tile_body<BF16, true>(p.dy, p.ymask, p.x, p.dw, p.db,
+    wgrad_tile_body<BF16, true>(p.dy, p.ymask, p.x, dw_out, db_out,
                                 p.db != nullptr, p.M, p.N, p.K, p.lddy,
-                                p.ldx, 0, p.M, bn0, bk0, smem, dbs);
+                                p.ldx, m_lo, m_hi, bn0, bk0, smem, dbs);
   else
-    wgrad_tile_body<BF16, false>(p.dy, p.ymask, p.x, p.dw, p.db,
+    wgrad_tile_body<BF16, false>(p.dy, p.ymask, p.x, dw_out, db_out,
                                  p.db != nullptr, p.M, p.N, p.K, p.lddy,
-                                 p.ldx, 0, p.M, bn0, bk0, smem, dbs);
+                                 p.ldx, m_lo, m_hi, bn0, bk0, smem, dbs);
+}
+
+// one deterministic combine for EVERY problem of the phase
+__global__ __launch_bounds__(256)
+void mwgrad_het_combine_kernel(WHArgs a, int split) {
+  const int64_t total = a.per_slab;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    float s = 0.f;
+    for (int sl = 0; sl < split; ++sl)
+      s += a.part[(int64_t)sl * a.per_slab + i];
+    int zi = 0;
+#pragma unroll
+    for (int z = 1; z < MAXW; ++z)
+      if (z < a.np && i >= a.p[z].poff) zi = z;
+    const WHProb& p = a.p[zi];
+    const int64_t off = i - p.poff;
+    const int64_t dwn = (int64_t)p.N * p.K;
+    if (off < dwn) p.dw[off] = s;
+    else if (p.db) p.db[off - dwn] = s;
+  }
 }
 
 // deterministic combine of split-M wgrad slabs: for each problem z,
@@ -1330,6 +1371,8 @@ void mwgrad_het(std::vector<torch::Tensor> dys,
   WHArgs a{};
   a.np = np;
   int blk = 0;
+  int64_t poff = 0;
+  int maxM = 0;
   for (int i = 0; i < np; ++i) {
     WHProb& p = a.p[i];
     p.dy = dys[i].data_ptr<float>();
@@ -1341,14 +1384,39 @@ void mwgrad_het(std::vector<torch::Tensor> dys,
     p.lddy = (int)lddys[i]; p.ldx = (int)ldxs[i];
     p.bx = (p.N + TB - 1) / TB;
     p.blk0 = blk;
+    p.poff = poff;
     blk += p.bx * ((p.K + TB - 1) / TB);
+    poff += (int64_t)p.N * p.K + p.N;
+    maxM = std::max(maxM, p.M);
   }
-  if (*g_bf16_flag)
-    hipLaunchKernelGGL((mwgrad_het_kernel<true>), dim3(blk), dim3(256), 0,
+  // split-M across blockIdx.z at large batch (mirrors mwgrad's split;
+  // ONE combine covers every problem of the phase)
+  const bool bf16 = *g_bf16_flag;
+  const int BKc = bf16 ? BKB2 : BKF2;
+  const int chunks = (maxM + BKc - 1) / BKc;
+  int split = std::max(1, std::min({chunks, (384 + blk - 1) / blk, 64}));
+  int m_chunk = ((maxM + split - 1) / split + BKc - 1) / BKc * BKc;
+  split = (maxM + m_chunk - 1) / m_chunk;
+  a.m_chunk = m_chunk;
+  a.per_slab = poff;
+  a.part = nullptr;
+  torch::Tensor part;
+  if (split > 1) {
+    part = torch::empty({(int64_t)split * poff, 1}, dws[0].options());
+    a.part = part.data_ptr<float>();
+  }
+  dim3 grid(blk, 1, split);
+  if (bf16)
+    hipLaunchKernelGGL((mwgrad_het_kernel<true>), grid, dim3(256), 0,
                        stream(), a);
   else
-    hipLaunchKernelGGL((mwgrad_het_kernel<false>), dim3(blk), dim3(256), 0,
+    hipLaunchKernelGGL((mwgrad_het_kernel<false>), grid, dim3(256), 0,
                        stream(), a);
+  if (split > 1) {
+    int blocks = (int)std::min<int64_t>((poff + 255) / 256, 768);
+    hipLaunchKernelGGL(mwgrad_het_combine_kernel, dim3(blocks), dim3(256),
+                       0, stream(), a, split);
+  }
 }
 
 void transpose_multi(std::vector<torch::Tensor> ws,

@@ -350,148 +350,6 @@ DEVINL void gemm_pass(const float* x, const float* w, const float* mask,
   }
 }
 
-// ---------------------------------------------------------------------------
-// Large-batch GEMM tile: 128x128 block, 64x64 wave tiles (4x4 16x16
-// frags), BK=64.  Doubles the MFMA work per staged LDS element vs the
-// 64x64 tile — the small-tile path measured VALU-staging-bound at
-// batch 4096 (~18 VALU insts per MFMA).  bf16 only; M>=2048, N>=128.
-// ---------------------------------------------------------------------------
-
-constexpr int LDS2 = 72;   // halves per row (64 + 8 pad, conflict-free)
-
-template <bool MASK>
-DEVINL void load_tile_regs128(float* v, const float* src,
-                              const float* mask, int r0, int k0, int R,
-                              int C, int ld) {
-  const int tid = threadIdx.x;
-  const int row = tid & 127;
-  const int gr = r0 + row;
-  const int c0 = (tid >> 7) * 32;
-  bool interior = (r0 + 128 <= R) && (k0 + 64 <= C) && ((ld & 3) == 0)
-                  && ((k0 & 3) == 0);
-  if (interior) {
-    const float* p = src + (int64_t)gr * ld + k0 + c0;
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      float4 f = *(const float4*)(p + q * 4);
-      v[q*4+0]=f.x; v[q*4+1]=f.y; v[q*4+2]=f.z; v[q*4+3]=f.w;
-    }
-    if constexpr (MASK) {
-      const float* mp = mask + (int64_t)gr * ld + k0 + c0;
-#pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        float4 f = *(const float4*)(mp + q * 4);
-        v[q*4+0] = f.x > 0.f ? v[q*4+0] : 0.f;
-        v[q*4+1] = f.y > 0.f ? v[q*4+1] : 0.f;
-        v[q*4+2] = f.z > 0.f ? v[q*4+2] : 0.f;
-        v[q*4+3] = f.w > 0.f ? v[q*4+3] : 0.f;
-      }
-    }
-  } else {
-#pragma unroll
-    for (int e = 0; e < 32; ++e) {
-      int c = k0 + c0 + e;
-      float val = 0.f;
-      if (gr < R && c < C) {
-        val = src[(int64_t)gr * ld + c];
-        if constexpr (MASK)
-          val = mask[(int64_t)gr * ld + c] > 0.f ? val : 0.f;
-      }
-      v[e] = val;
-    }
-  }
-}
-
-DEVINL void write_tile_lds128(void* lds, const float* v) {
-  const int tid = threadIdx.x;
-  const int row = tid & 127;
-  const int c0 = (tid >> 7) * 32;
-  __bf16* d = (__bf16*)lds;
-  union { __bf16 h[32]; uint4 u[4]; } pk;
-#pragma unroll
-  for (int e = 0; e < 32; ++e) pk.h[e] = (__bf16)v[e];
-  uint4* dst = (uint4*)&d[row * LDS2 + c0];
-#pragma unroll
-  for (int q = 0; q < 4; ++q) dst[q] = pk.u[q];
-}
-
-DEVINL void mma_tiles_big(const void* xs_, const void* ws_,
-                          f32x4 (&acc)[4][4], int lane, int wrow,
-                          int wcol) {
-  const __bf16* xs = (const __bf16*)xs_;
-  const __bf16* ws = (const __bf16*)ws_;
-  const int arow = lane & 15;
-  const int ak0 = (lane >> 4) * 8;
-#pragma unroll
-  for (int kk = 0; kk < 64; kk += 32) {
-    bf16x8 a[4], b[4];
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-      a[mi] = *(const bf16x8*)&xs[(wrow + mi * 16 + arow) * LDS2
-                                  + kk + ak0];
-#pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
-      b[ni] = *(const bf16x8*)&ws[(wcol + ni * 16 + arow) * LDS2
-                                  + kk + ak0];
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a[mi], b[ni], acc[mi][ni], 0, 0, 0);
-  }
-}
-
-template <bool MASK, bool RELU>
-__global__ __launch_bounds__(256)
-void mgemm_big_kernel(MGemm g) {
-  const MProb& p = g.p[blockIdx.z];
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int wrow = (wid >> 1) * 64;
-  const int wcol = (wid & 1) * 64;
-  const int bm0 = blockIdx.x * 128;
-  const int bn0 = blockIdx.y * 128;
-  constexpr int LB = 128 * LDS2 * 2;
-  __shared__ __attribute__((aligned(16))) char smem[2 * LB];
-  void* xs = smem;
-  void* ws = smem + LB;
-  f32x4 acc[4][4] = {};
-  float va[32], vb[32];
-  load_tile_regs128<MASK>(va, p.x, p.mask, bm0, 0, g.M, g.K, g.lda);
-  load_tile_regs128<false>(vb, p.w, nullptr, bn0, 0, g.N, g.K, g.K);
-  for (int k0 = 0; k0 < g.K; k0 += 64) {
-    write_tile_lds128(xs, va);
-    write_tile_lds128(ws, vb);
-    if (k0 + 64 < g.K) {
-      load_tile_regs128<MASK>(va, p.x, p.mask, bm0, k0 + 64, g.M, g.K,
-                              g.lda);
-      load_tile_regs128<false>(vb, p.w, nullptr, bn0, k0 + 64, g.N,
-                               g.K, g.K);
-    }
-    __syncthreads();
-    mma_tiles_big(xs, ws, acc, lane, wrow, wcol);
-    __syncthreads();
-  }
-  const int crow = (lane >> 4) * 4, ccol = lane & 15;
-#pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int grow = bm0 + wrow + mi * 16 + crow + r;
-        int gcol = bn0 + wcol + ni * 16 + ccol;
-        if (grow < g.M && gcol < g.N) {
-          float v = acc[mi][ni][r];
-          if (p.bias) v += p.bias[gcol];
-          if constexpr (RELU) v = fmaxf(v, 0.f);
-          p.y[(int64_t)grow * g.ldy + gcol] = v;
-        }
-      }
-}
-
 template <bool BF16, bool MASK, bool RELU, bool SUM2>
 __global__ __launch_bounds__(256)
 void mgemm_kernel(MGemm g) {
@@ -1416,20 +1274,6 @@ void mgemm(std::vector<torch::Tensor> xs, std::vector<torch::Tensor> ws,
   if (split > 1) {
     part = torch::empty({(int64_t)split * nz, M * N}, ys[0].options());
     g.part = part.data_ptr<float>();
-  }
-  // big-tile path for throughput shapes (staging-VALU-bound otherwise)
-  if (bf16 && !sum2 && split == 1 && M >= 2048 && N >= 128) {
-    dim3 bgrid((M + 127) / 128, (N + 127) / 128, nz);
-    auto LB = [&](auto m, auto r) {
-      hipLaunchKernelGGL((mgemm_big_kernel<decltype(m)::value,
-                                           decltype(r)::value>),
-                         bgrid, dim3(256), 0, stream(), g);
-    };
-    if (has_mask) { if (relu) LB(std::true_type{}, std::true_type{});
-                    else LB(std::true_type{}, std::false_type{}); }
-    else          { if (relu) LB(std::false_type{}, std::true_type{});
-                    else LB(std::false_type{}, std::false_type{}); }
-    return;
   }
   dim3 grid(gx, gy, nz * split);
   auto L = [&](auto b, auto m, auto r, auto s) {

@@ -229,6 +229,15 @@ def mlp_forward(x, layers, relu_last: bool = True):
 # trunk's shape family, reference networks/convolutional.py:30-51)
 # ---------------------------------------------------------------------------
 
+
+def _conv_mask_in_kernel() -> bool:
+    """TAC_AMD_CONV_MASK=1: apply the relu mask inside the conv
+    dgrad/wgrad gathers instead of a threshold_backward launch per dy
+    (re-A/B'd after the stride-class dgrad)."""
+    import os
+    return os.environ.get("TAC_AMD_CONV_MASK") == "1"
+
+
 class _NativeConv2d(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride, relu):
@@ -249,25 +258,29 @@ class _NativeConv2d(torch.autograd.Function):
     def backward(ctx, dy):
         ext = require_extension()
         dy = dy.contiguous()
+        m = None
         if ctx.relu:
-            # one elementwise relu-backward beats masked gathers inside
-            # the conv kernels (measured)
             x, w, y, b = ctx.saved_tensors
-            dy = torch.ops.aten.threshold_backward(dy, y, 0)
+            if _conv_mask_in_kernel():
+                m = y
+            else:
+                # one elementwise relu-backward instead of masked
+                # gathers inside the conv kernels (A/B, default)
+                dy = torch.ops.aten.threshold_backward(dy, y, 0)
         else:
             x, w, b = ctx.saved_tensors
         dx = None
         if ctx.needs_input_grad[0]:
             wt = _conv_wt(ext, w)
-            dx = ext.conv2d_dgrad(dy, None, wt, x, w, ctx.stride)
+            dx = ext.conv2d_dgrad(dy, m, wt, x, w, ctx.stride)
         dw = db = None
         if ctx.needs_input_grad[1]:
             outs = _direct_outs([(w, b)]) if ctx.has_bias else None
             if outs is not None:
-                ext.conv2d_wgrad_multi([dy], [None], [x], w, ctx.stride,
+                ext.conv2d_wgrad_multi([dy], [m], [x], w, ctx.stride,
                                        out=[outs[0][0], outs[0][1]])
             else:
-                dw, db = ext.conv2d_wgrad(dy, None, x, w, ctx.stride)
+                dw, db = ext.conv2d_wgrad(dy, m, x, w, ctx.stride)
                 if not ctx.has_bias:
                     db = None
         return dx, dw, db, None, None
@@ -385,17 +398,21 @@ class _PairedConv2d(torch.autograd.Function):
         ext = require_extension()
         dy1 = dy1.contiguous()
         dy2 = dy2.contiguous()
+        m1 = m2 = None
         if ctx.relu:
             x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
-            dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
-            dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
+            if _conv_mask_in_kernel():
+                m1, m2 = y1, y2
+            else:
+                dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
+                dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
         else:
             x1, x2, w1, w2, b1, b2 = ctx.saved_tensors
         dx1 = dx2 = None
         if ctx.needs_input_grad[0]:
             wt1 = _conv_wt(ext, w1)
             wt2 = _conv_wt(ext, w2)
-            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [None, None],
+            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [m1, m2],
                                               [wt1, wt2], x1, w1,
                                               ctx.stride)
         dw1 = db1 = dw2 = db2 = None
@@ -404,11 +421,11 @@ class _PairedConv2d(torch.autograd.Function):
                     else None)
             if outs is not None:
                 ext.conv2d_wgrad_multi(
-                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride,
+                    [dy1, dy2], [m1, m2], [x1, x2], w1, ctx.stride,
                     out=[outs[0][0], outs[0][1], outs[1][0], outs[1][1]])
             else:
                 dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
-                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
+                    [dy1, dy2], [m1, m2], [x1, x2], w1, ctx.stride)
                 if not ctx.has_bias:
                     db1 = db2 = None
         return dx1, dx2, dw1, db1, dw2, db2, None, None
@@ -531,17 +548,21 @@ class _QuadConv2d(torch.autograd.Function):
         ext = require_extension()
         dy1 = dy1.contiguous()
         dy2 = dy2.contiguous()
+        m1 = m2 = None
         if ctx.relu:
             x1, x2, w1, w2, y1, y2, b1, b2 = ctx.saved_tensors
-            dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
-            dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
+            if _conv_mask_in_kernel():
+                m1, m2 = y1, y2
+            else:
+                dy1 = torch.ops.aten.threshold_backward(dy1, y1, 0)
+                dy2 = torch.ops.aten.threshold_backward(dy2, y2, 0)
         else:
             x1, x2, w1, w2, b1, b2 = ctx.saved_tensors
         dx1 = dx2 = None
         if ctx.needs_input_grad[2]:
             wt1 = _conv_wt(ext, w1)
             wt2 = _conv_wt(ext, w2)
-            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [None, None],
+            dx1, dx2 = ext.conv2d_dgrad_multi([dy1, dy2], [m1, m2],
                                               [wt1, wt2], x1, w1,
                                               ctx.stride)
         dw1 = db1 = dw2 = db2 = None
@@ -549,11 +570,11 @@ class _QuadConv2d(torch.autograd.Function):
             outs = _direct_outs([(w1, b1), (w2, b2)])
             if outs is not None:
                 ext.conv2d_wgrad_multi(
-                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride,
+                    [dy1, dy2], [m1, m2], [x1, x2], w1, ctx.stride,
                     out=[outs[0][0], outs[0][1], outs[1][0], outs[1][1]])
             else:
                 dw1, db1, dw2, db2 = ext.conv2d_wgrad_multi(
-                    [dy1, dy2], [None, None], [x1, x2], w1, ctx.stride)
+                    [dy1, dy2], [m1, m2], [x1, x2], w1, ctx.stride)
         return (None, None, dx1, dx2, None, None, None, None,
                 dw1, db1, dw2, db2, None, None)
 

@@ -60,6 +60,14 @@ class _SyntheticVisualEnv(Env):
             torch.from_numpy(self._render_frame(feats)),
         )
 
+    def process_observations(self, feats) -> MultiObservation:
+        """Build a MultiObservation from raw per-step features (the
+        reference's method name, environments/wall_runner.py:38-58 —
+        there it concatenates dm_control sensor groups and rolls the
+        egocentric camera; here the synthetic dynamics provide the
+        feature vector directly)."""
+        return self._obs(np.asarray(feats, dtype=np.float32))
+
     def reset(self):
         return self._obs(self._inner.reset())
 

@@ -21,6 +21,14 @@ class StateNormalizer(abc.ABC):
     def save(self, artifact_path: str = "normalizer"):
         ckpt.log_state_dict(self.state_dict(), artifact_path)
 
+    # reference API names (sac/utils.py:19-24): save_state logs the
+    # state dict to an artifact path, load_state restores from one
+    def save_state(self, path: str = "normalizer"):
+        self.save(path)
+
+    def load_state(self, state_dict: dict):
+        self.load_state_dict(state_dict)
+
     def state_dict(self) -> dict:
         return {}
 

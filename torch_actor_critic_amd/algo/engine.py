@@ -2,7 +2,7 @@
 
 Replaces the autograd-recorded update (112 kernels/update, 75% of GPU
 time in scalar-staged GEMMs — profiles/r01_baseline_update_profile.md)
-with an explicit kernel schedule of ~40 launches built from the
+with an explicit kernel schedule of 26 launches built from the
 multi-problem MFMA kernels in ops/csrc/fused.hip:
 
  * one Philox counter bump covers the replay draw AND the policy noise;

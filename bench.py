@@ -170,12 +170,16 @@ def main():
     state = env.reset()
 
     act_graph = None
+    vis_act = None
     if on_gpu and not visual:
         from torch_actor_critic_amd.algo.act import (WindowedStore,
                                                      make_act_path)
         act_graph = make_act_path(actor, obs_dim, act_dim, device,
                                   4321 + 7919 * rank)
         wstore = WindowedStore(buffer, args.update_window)
+    elif on_gpu and visual:
+        from torch_actor_critic_amd.algo.act import VisualActGraph
+        vis_act = VisualActGraph(actor, obs_dim, vis_dim, act_dim, device)
 
     do_act = args.mode in ("full", "acting")
     do_upd = args.mode in ("full", "updates")
@@ -197,15 +201,20 @@ def main():
                     MultiObservation)
                 with torch.no_grad():
                     for _ in range(n_steps):
-                        if isinstance(state, MultiObservation):
-                            s = MultiObservation(state.features.to(device),
-                                                 state.frame.to(device))
+                        if vis_act is not None:
+                            a_np = vis_act.act(state)
                         else:
-                            s = torch.as_tensor(state, dtype=torch.float32,
-                                                device=device)
-                        a, _ = actor(s, deterministic=False,
-                                     with_logprob=False)
-                        a_np = a.detach().cpu().numpy()
+                            if isinstance(state, MultiObservation):
+                                s = MultiObservation(
+                                    state.features.to(device),
+                                    state.frame.to(device))
+                            else:
+                                s = torch.as_tensor(state,
+                                                    dtype=torch.float32,
+                                                    device=device)
+                            a, _ = actor(s, deterministic=False,
+                                         with_logprob=False)
+                            a_np = a.detach().cpu().numpy()
                         nstate, reward, done, _ = env.step(a_np)
                         buffer.store(state, a_np, float(reward), nstate,
                                      float(done))

@@ -166,3 +166,47 @@ class WindowedStore:
         self.buffer.store_batch(self.obs[:n], self.act[:n], self.rew[:n],
                                 self.nobs[:n], self.done[:n])
         self.n = 0
+
+
+class VisualActGraph:
+    """hipGraph-captured single-state stochastic VISUAL actor forward
+    (conv trunk + dual-stream tanh-Gaussian head) with pinned-staging
+    I/O — replaces ~0.4 ms of eager per-step acting with two H2D
+    copies + one graph replay + one D2H copy.  Philox noise is drawn by
+    a captured device-counter kernel, so every replay acts with fresh
+    noise (same mechanism as the update graphs)."""
+
+    def __init__(self, actor, feat_dim: int, vis_dim, act_dim: int,
+                 device: torch.device, warmup: int = 3):
+        from ..envs.visual import MultiObservation
+        self.device = device
+        self.feat_in = torch.zeros(feat_dim, device=device)
+        self.frame_in = torch.zeros(*vis_dim, device=device)
+        self.feat_pin = torch.zeros(feat_dim, pin_memory=True)
+        self.frame_pin = torch.zeros(*vis_dim, pin_memory=True)
+        self.act_pin = torch.zeros(act_dim, pin_memory=True)
+        self.ev = torch.cuda.Event()
+        mo = MultiObservation(self.feat_in, self.frame_in)
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(warmup):
+                actor(mo, deterministic=False, with_logprob=False)
+        torch.cuda.current_stream().wait_stream(s)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph), torch.no_grad():
+            self.act_out, _ = actor(mo, deterministic=False,
+                                    with_logprob=False)
+
+    def act(self, state) -> np.ndarray:
+        self.feat_pin.copy_(state.features.reshape(-1))
+        self.frame_pin.copy_(state.frame)
+        self.feat_in.copy_(self.feat_pin, non_blocking=True)
+        self.frame_in.copy_(self.frame_pin, non_blocking=True)
+        self.graph.replay()
+        self.act_pin.copy_(self.act_out.reshape(-1), non_blocking=True)
+        self.ev.record()
+        self.ev.synchronize()
+        return self.act_pin.numpy().copy()

@@ -346,6 +346,7 @@ class SAC:
         # stores (GPU, vector-obs envs only; visual envs use eager acting)
         act_graph = None
         wstore = None
+        _tried_visual_act = False
         if (device.type == "cuda" and self.use_graph
                 and hasattr(buffer, "obs_dim")):
             try:
@@ -388,10 +389,28 @@ class SAC:
                     state = self.normalizer.normalize_state(st).numpy()
                 if step < self.start_steps:
                     action = env.action_space.sample()
-                elif act_graph is not None:
-                    action = act_graph.act(state)
                 else:
-                    action = self._select_action(actor, state, device)
+                    if (act_graph is None and not _tried_visual_act
+                            and device.type == "cuda" and self.use_graph
+                            and isinstance(state, MultiObservation)):
+                        # lazily capture the visual act graph — the obs
+                        # dims come from the first MultiObservation
+                        _tried_visual_act = True
+                        try:
+                            from .act import VisualActGraph
+                            act_graph = VisualActGraph(
+                                actor, int(state.features.numel()),
+                                tuple(state.frame.shape),
+                                int(np.prod(env.action_space.shape)),
+                                device)
+                        except Exception as e:  # noqa: BLE001
+                            logger.warning(
+                                "visual act-graph capture failed (%r); "
+                                "using eager acting", e)
+                    if act_graph is not None:
+                        action = act_graph.act(state)
+                    else:
+                        action = self._select_action(actor, state, device)
 
                 next_state, reward, done, _info = env.step(action)
                 thr.tick_env()

@@ -338,3 +338,51 @@ def test_quad_forward_with_target_matches_sequential():
     for g, r in zip(got, ref_grads):
         assert torch.allclose(g, r, atol=1e-3, rtol=1e-3), \
             (g - r).abs().max()
+
+
+def test_visual_act_graph():
+    """Captured B=1 visual acting: correct shape/range, fresh Philox
+    noise per replay, and the sac.train lazy path completes a short
+    visual training run with policy actions."""
+    import torch
+    from torch_actor_critic_amd.algo.act import VisualActGraph
+    from torch_actor_critic_amd.envs.visual import MultiObservation
+    from torch_actor_critic_amd.models.visual import VisualActor
+
+    torch.manual_seed(3)
+    dev = torch.device(DEV)
+    actor = VisualActor(16, 4, (3, 64, 64), [32, 32],
+                        act_limit=1.0).to(dev)
+    vag = VisualActGraph(actor, 16, (3, 64, 64), 4, dev)
+    mo = MultiObservation(torch.randn(16), torch.randn(3, 64, 64))
+    a1 = vag.act(mo)
+    a2 = vag.act(mo)
+    assert a1.shape == (4,)
+    assert np.all(np.abs(a1) <= 1.0 + 1e-6)
+    assert not np.allclose(a1, a2)          # fresh noise per replay
+    # different observation influences the action distribution
+    mo2 = MultiObservation(torch.randn(16) * 3, torch.rand(3, 64, 64))
+    a3 = vag.act(mo2)
+    assert np.isfinite(a3).all()
+
+    # sac.train exercises the lazy capture (start_steps=0 -> policy acts)
+    from torch_actor_critic_amd.algo.sac import SAC
+    from torch_actor_critic_amd.buffer.visual import VisualReplayBuffer
+    from torch_actor_critic_amd.models.visual import VisualDoubleCritic
+    from torch_actor_critic_amd.optim import FlatAdam
+    from torch_actor_critic_amd import envs
+
+    env = envs.make("DeepMindWallRunner-v0")
+    env.seed(0)
+    actor2 = VisualActor(168, 56, (3, 64, 64), [32, 32],
+                         act_limit=1.0).to(dev)
+    critic2 = VisualDoubleCritic(168, 56, (3, 64, 64), [32, 32]).to(dev)
+    buf = VisualReplayBuffer(2000, 56, device=dev)
+    sac = SAC(alpha=0.2, gamma=0.99, polyak=0.995, reward_scale=1.0,
+              epochs=1, batch_size=8, start_steps=0, steps_per_epoch=60,
+              max_ep_len=50, update_after=30, update_every=10,
+              save_every=10**9)
+    sac.train(0, env, actor2, critic2, buf,
+              FlatAdam(actor2), FlatAdam(critic2), render=False,
+              logging=False)
+    assert buf.size >= 60

@@ -51,3 +51,7 @@ with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
     torch.cuda.synchronize()
 Fo.set_graph_opt(None, False)
 print(prof.key_averages().table(sort_by="cuda_time_total", row_limit=28, max_name_column_width=46))
+print("\n== aten ops only ==")
+for ev in sorted(prof.key_averages(), key=lambda e: -e.self_device_time_total):
+    if ev.key.startswith("aten::") and ev.self_device_time_total > 0:
+        print(f"{ev.self_device_time_total/1000:8.2f}ms {ev.count:5d}  {ev.key}")

@@ -16,9 +16,11 @@ Graph-replay safety is designed in everywhere:
  * all parameters/gradients are stable flat buffers (parallel/flat.py);
  * losses accumulate into static device tensors — zero host sync.
 
-Data-parallel mode (world>1) splits the capture into three graphs with
-the two flat-bucket RCCL all-reduces between them (collective capture is
-backend-dependent; the split costs two eager launches per update).
+Data-parallel mode (world>1) records BOTH flat-bucket RCCL all-reduces
+INSIDE the one captured graph (RCCL supports captured collectives) — a
+DP update is a single replay; if the communicator refuses capture the
+module falls back to three graphs with host-issued collectives between
+them.
 """
 
 import logging

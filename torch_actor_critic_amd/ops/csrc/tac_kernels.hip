@@ -1057,6 +1057,93 @@ void replay_sample_into(torch::Tensor state, torch::Tensor act,
                      od.data_ptr<float>(), obs_dim, act_dim);
 }
 
+// One-kernel visual ring store: quantize both fp32 frames to u8 and
+// write every field of the transition at ring slot `idx` — replaces the
+// CPU-side encode chain + ~10 aten copies per env step.
+template <bool QUANT>
+__global__ __launch_bounds__(256)
+void visual_store_kernel(const float* __restrict__ feat,
+                         const float* __restrict__ frame,
+                         const float* __restrict__ nfeat,
+                         const float* __restrict__ nframe,
+                         const float* __restrict__ act,
+                         float rew, float done,
+                         float* __restrict__ rfeat,
+                         void* __restrict__ rframe,
+                         float* __restrict__ rnfeat,
+                         void* __restrict__ rnframe,
+                         float* __restrict__ ract,
+                         float* __restrict__ rrew,
+                         float* __restrict__ rdone,
+                         int64_t idx, int feat_dim, int64_t frame_n,
+                         int act_dim) {
+  const int slice = blockIdx.x;
+  const int nslice = gridDim.x;
+  if (slice == 0) {
+    for (int c = threadIdx.x; c < feat_dim; c += blockDim.x) {
+      rfeat[idx * feat_dim + c] = feat[c];
+      rnfeat[idx * feat_dim + c] = nfeat[c];
+    }
+    for (int c = threadIdx.x; c < act_dim; c += blockDim.x)
+      ract[idx * act_dim + c] = act[c];
+    if (threadIdx.x == 0) {
+      rrew[idx] = rew;
+      rdone[idx] = done;
+    }
+  }
+  const int64_t chunk = (frame_n + nslice - 1) / nslice;
+  const int64_t lo = (int64_t)slice * chunk;
+  const int64_t hi = min(frame_n, lo + chunk);
+  if constexpr (QUANT) {
+    uint8_t* f0 = (uint8_t*)rframe + idx * frame_n;
+    uint8_t* f1 = (uint8_t*)rnframe + idx * frame_n;
+    for (int64_t c = lo + threadIdx.x; c < hi; c += blockDim.x) {
+      // matches the host encode: round((clamp(v,-1,1)+1)*127.5)
+      float v0 = fminf(fmaxf(frame[c], -1.f), 1.f);
+      float v1 = fminf(fmaxf(nframe[c], -1.f), 1.f);
+      f0[c] = (uint8_t)lrintf((v0 + 1.f) * 127.5f);
+      f1[c] = (uint8_t)lrintf((v1 + 1.f) * 127.5f);
+    }
+  } else {
+    float* f0 = (float*)rframe + idx * frame_n;
+    float* f1 = (float*)rnframe + idx * frame_n;
+    for (int64_t c = lo + threadIdx.x; c < hi; c += blockDim.x) {
+      f0[c] = frame[c];
+      f1[c] = nframe[c];
+    }
+  }
+}
+
+void visual_store_into(torch::Tensor feat, torch::Tensor frame,
+                       torch::Tensor nfeat, torch::Tensor nframe,
+                       torch::Tensor act, double rew, double done,
+                       torch::Tensor rfeat, torch::Tensor rframe,
+                       torch::Tensor rnfeat, torch::Tensor rnframe,
+                       torch::Tensor ract, torch::Tensor rrew,
+                       torch::Tensor rdone, int64_t idx) {
+  CHECK_IN(feat); CHECK_IN(rfeat);
+  const int feat_dim = (int)feat.numel();
+  const int act_dim = (int)act.numel();
+  const int64_t frame_n = frame.numel();
+  const bool quant = rframe.scalar_type() == torch::kUInt8;
+  const int slices = (int)std::min<int64_t>(8, (frame_n + 4095) / 4096);
+  auto s = cur_stream();
+  auto L = [&](auto q) {
+    hipLaunchKernelGGL((visual_store_kernel<decltype(q)::value>),
+                       dim3(std::max(1, slices)), dim3(256), 0, s,
+                       feat.data_ptr<float>(), frame.data_ptr<float>(),
+                       nfeat.data_ptr<float>(), nframe.data_ptr<float>(),
+                       act.data_ptr<float>(), (float)rew, (float)done,
+                       rfeat.data_ptr<float>(), rframe.data_ptr(),
+                       rnfeat.data_ptr<float>(), rnframe.data_ptr(),
+                       ract.data_ptr<float>(), rrew.data_ptr<float>(),
+                       rdone.data_ptr<float>(), idx, feat_dim, frame_n,
+                       act_dim);
+  };
+  if (quant) L(std::true_type{});
+  else L(std::false_type{});
+}
+
 void visual_sample_into(torch::Tensor feat, torch::Tensor frames,
                         torch::Tensor nfeat, torch::Tensor nframes,
                         torch::Tensor act, torch::Tensor rew,
@@ -1169,6 +1256,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("replay_sample", &replay_sample);
   m.def("replay_sample_into", &replay_sample_into);
   m.def("visual_sample_into", &visual_sample_into);
+  m.def("visual_store_into", &visual_store_into);
   m.def("philox_randn_", &philox_randn_);
   m.def("bump_counter", &bump_counter);
 }

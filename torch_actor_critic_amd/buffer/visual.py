@@ -118,15 +118,28 @@ class VisualReplayBuffer:
                 "a": torch.empty(ad, **dev),
             }
         st = self._stage
-        st["pf"].copy_(obs.features.reshape(-1))
-        st["pF"].copy_(obs.frame)
+        # obs is usually last step's next_obs (state = nstate in the env
+        # loop): ping-pong the staged buffers instead of re-staging —
+        # saves one pinned CPU copy + one H2D of the 84 KB frame per step
+        last = getattr(self, "_last_next_src", None)
+        if (last is not None and obs.features is last[0]
+                and obs.frame is last[1]):
+            st["f"], st["nf"] = st["nf"], st["f"]
+            st["F"], st["nF"] = st["nF"], st["F"]
+            st["pf"], st["pnf"] = st["pnf"], st["pf"]
+            st["pF"], st["pnF"] = st["pnF"], st["pF"]
+        else:
+            st["pf"].copy_(obs.features.reshape(-1))
+            st["pF"].copy_(obs.frame)
+            st["f"].copy_(st["pf"], non_blocking=True)
+            st["F"].copy_(st["pF"], non_blocking=True)
         st["pnf"].copy_(next_obs.features.reshape(-1))
         st["pnF"].copy_(next_obs.frame)
         st["pa"].copy_(torch.as_tensor(np.asarray(act),
                                        dtype=torch.float32).reshape(-1))
-        for d, p in (("f", "pf"), ("F", "pF"), ("nf", "pnf"),
-                     ("nF", "pnF"), ("a", "pa")):
+        for d, p in (("nf", "pnf"), ("nF", "pnF"), ("a", "pa")):
             st[d].copy_(st[p], non_blocking=True)
+        self._last_next_src = (next_obs.features, next_obs.frame)
         ext.visual_store_into(st["f"], st["F"], st["nf"], st["nF"],
                               st["a"], float(rew), float(done),
                               self.features, self.frames,

@@ -202,7 +202,7 @@ def main():
                 with torch.no_grad():
                     for _ in range(n_steps):
                         if vis_act is not None:
-                            a_np = vis_act.act(state)
+                            a_np = vis_act.act(state, buffer=buffer)
                         else:
                             if isinstance(state, MultiObservation):
                                 s = MultiObservation(

@@ -200,11 +200,26 @@ class VisualActGraph:
             self.act_out, _ = actor(mo, deterministic=False,
                                     with_logprob=False)
 
-    def act(self, state) -> np.ndarray:
-        self.feat_pin.copy_(state.features.reshape(-1))
-        self.frame_pin.copy_(state.frame)
-        self.feat_in.copy_(self.feat_pin, non_blocking=True)
-        self.frame_in.copy_(self.frame_pin, non_blocking=True)
+    def act(self, state, buffer=None) -> np.ndarray:
+        # when the buffer's fused store already staged this observation
+        # on device (state is last step's next_obs), read it D2D instead
+        # of re-uploading the 84 KB frame through pinned memory
+        staged = None
+        if buffer is not None:
+            last = getattr(buffer, "_last_next_src", None)
+            st = getattr(buffer, "_stage", None)
+            if (last is not None and st is not None
+                    and state.features is last[0]
+                    and state.frame is last[1]):
+                staged = (st["nf"], st["nF"])
+        if staged is not None:
+            self.feat_in.copy_(staged[0], non_blocking=True)
+            self.frame_in.copy_(staged[1], non_blocking=True)
+        else:
+            self.feat_pin.copy_(state.features.reshape(-1))
+            self.frame_pin.copy_(state.frame)
+            self.feat_in.copy_(self.feat_pin, non_blocking=True)
+            self.frame_in.copy_(self.frame_pin, non_blocking=True)
         self.graph.replay()
         self.act_pin.copy_(self.act_out.reshape(-1), non_blocking=True)
         self.ev.record()

@@ -408,7 +408,10 @@ class SAC:
                                 "visual act-graph capture failed (%r); "
                                 "using eager acting", e)
                     if act_graph is not None:
-                        action = act_graph.act(state)
+                        if isinstance(state, MultiObservation):
+                            action = act_graph.act(state, buffer=buffer)
+                        else:
+                            action = act_graph.act(state)
                     else:
                         action = self._select_action(actor, state, device)
 

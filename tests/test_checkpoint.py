@@ -91,3 +91,32 @@ def test_resume_via_main_load_session(tmp_path, monkeypatch):
     o1, _ = actor(obs, deterministic=True)
     o2, _ = a2(obs, deterministic=True)
     assert torch.allclose(o1, o2)
+
+
+def test_visual_model_roundtrip(tmp_path):
+    """Visual actor/critic modules survive the mlruns save/load cycle
+    (pickled nn.Modules, reference mlflow.pytorch.log_model layout)."""
+    import torch as _t
+    from networks.convolutional import VisualActor, VisualDoubleCritic
+    from torch_actor_critic_amd.envs.visual import MultiObservation
+
+    _setup(tmp_path)
+    _t.manual_seed(2)
+    actor = VisualActor(12, 4, (3, 32, 32), [16, 16], act_limit=1.0)
+    critic = VisualDoubleCritic(12, 4, (3, 32, 32), [16, 16])
+    ckpt.log_model(actor, "actor")
+    ckpt.log_model(critic, "critic")
+    run_id = ckpt.active_run_id()
+    ckpt.end_run()
+    base = str(tmp_path / "mlruns" / "0" / run_id / "artifacts")
+    a2 = ckpt.load_model(base + "/actor")
+    c2 = ckpt.load_model(base + "/critic")
+    mo = MultiObservation(_t.randn(12), _t.randn(3, 32, 32))
+    with _t.no_grad():
+        p1, _ = actor(mo, deterministic=True)
+        p2, _ = a2(mo, deterministic=True)
+        q1 = critic(MultiObservation(_t.randn(2, 12), _t.randn(2, 3, 32, 32)),
+                    _t.rand(2, 4))
+    assert _t.allclose(p1, p2)
+    for pa, pb in zip(critic.parameters(), c2.parameters()):
+        assert _t.equal(pa, pb)

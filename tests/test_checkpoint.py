@@ -102,8 +102,8 @@ def test_visual_model_roundtrip(tmp_path):
 
     _setup(tmp_path)
     _t.manual_seed(2)
-    actor = VisualActor(12, 4, (3, 32, 32), [16, 16], act_limit=1.0)
-    critic = VisualDoubleCritic(12, 4, (3, 32, 32), [16, 16])
+    actor = VisualActor(12, 4, (3, 64, 64), [16, 16], act_limit=1.0)
+    critic = VisualDoubleCritic(12, 4, (3, 64, 64), [16, 16])
     ckpt.log_model(actor, "actor")
     ckpt.log_model(critic, "critic")
     run_id = ckpt.active_run_id()
@@ -111,11 +111,11 @@ def test_visual_model_roundtrip(tmp_path):
     base = str(tmp_path / "mlruns" / "0" / run_id / "artifacts")
     a2 = ckpt.load_model(base + "/actor")
     c2 = ckpt.load_model(base + "/critic")
-    mo = MultiObservation(_t.randn(12), _t.randn(3, 32, 32))
+    mo = MultiObservation(_t.randn(12), _t.randn(3, 64, 64))
     with _t.no_grad():
         p1, _ = actor(mo, deterministic=True)
         p2, _ = a2(mo, deterministic=True)
-        q1 = critic(MultiObservation(_t.randn(2, 12), _t.randn(2, 3, 32, 32)),
+        q1 = critic(MultiObservation(_t.randn(2, 12), _t.randn(2, 3, 64, 64)),
                     _t.rand(2, 4))
     assert _t.allclose(p1, p2)
     for pa, pb in zip(critic.parameters(), c2.parameters()):
